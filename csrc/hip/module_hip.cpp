@@ -1,0 +1,180 @@
+// pybind11 host module for the gfx950 HIP kernels (`kungfu_amd._hip`).
+//
+// No torch headers and no hipify: tensors cross as raw device pointers
+// (tensor.data_ptr()) and HIP streams as integers
+// (torch.cuda.current_stream().cuda_stream); the Python wrapper in
+// kungfu_amd/ops/hip.py owns dtype/shape checking.
+#include <hip/hip_runtime.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+#include <vector>
+
+namespace py = pybind11;
+
+extern "C" {
+hipError_t kf_pack(const void *, int, void *, int, void *);
+hipError_t kf_unpack(const void *, int, const void *, float, int, void *);
+hipError_t kf_avg_inplace(void *, const void *, float, long long, int,
+                          void *);
+hipError_t kf_scale(void *, float, long long, int, void *);
+hipError_t kf_norm2(const void *, long long, void *, int, void *);
+hipError_t kf_dot(const void *, const void *, long long, void *, int,
+                  void *);
+hipError_t kf_sgd_momentum(void *, const void *, void *, long long, float,
+                           float, float, float, int, int, void *);
+hipError_t kf_transform2(void *, const void *, long long, int, int, void *);
+}
+
+namespace {
+
+void check(hipError_t e, const char *what)
+{
+    if (e != hipSuccess) {
+        throw std::runtime_error(std::string(what) + ": " +
+                                 hipGetErrorString(e));
+    }
+}
+
+struct ChunkHost {
+    const void *src;
+    unsigned long long dst;
+    unsigned int n;
+};
+
+// Precomputed device-side chunk table for fusion pack/unpack of one bucket.
+// segments: list of (device_ptr, fused_elem_offset, numel). The table splits
+// segments into <=chunk_elems pieces so blocks load-balance (SURVEY §2.6
+// item 6: ResNet-50 has 161 tensors from 1 to 2.3M elements).
+class FusionPlan {
+  public:
+    FusionPlan(const std::vector<std::tuple<uintptr_t, uint64_t, uint64_t>>
+                   &segments,
+               int dtype, unsigned chunk_elems = 1u << 14)
+        : dtype_(dtype)
+    {
+        std::vector<ChunkHost> chunks;
+        total_ = 0;
+        for (const auto &seg : segments) {
+            const uintptr_t ptr = std::get<0>(seg);
+            const uint64_t off = std::get<1>(seg);
+            uint64_t n = std::get<2>(seg);
+            total_ += n;
+            uint64_t done = 0;
+            const size_t esize = dtype == 0 ? 4 : 2;
+            while (done < n) {
+                const uint64_t take = std::min<uint64_t>(chunk_elems,
+                                                         n - done);
+                ChunkHost c;
+                c.src = (const void *)(ptr + done * esize);
+                c.dst = off + done;
+                c.n = (unsigned int)take;
+                chunks.push_back(c);
+                done += take;
+            }
+        }
+        nchunks_ = (int)chunks.size();
+        if (nchunks_ > 0) {
+            check(hipMalloc(&dev_, sizeof(ChunkHost) * nchunks_),
+                  "hipMalloc(FusionPlan)");
+            check(hipMemcpy(dev_, chunks.data(),
+                            sizeof(ChunkHost) * nchunks_,
+                            hipMemcpyHostToDevice),
+                  "hipMemcpy(FusionPlan)");
+        }
+    }
+    ~FusionPlan()
+    {
+        if (dev_) (void)hipFree(dev_);
+    }
+    FusionPlan(const FusionPlan &) = delete;
+
+    void pack(uintptr_t fused, uintptr_t stream)
+    {
+        check(kf_pack(dev_, nchunks_, (void *)fused, dtype_,
+                      (void *)stream),
+              "kf_pack");
+    }
+    void unpack(uintptr_t fused, float scale, uintptr_t stream)
+    {
+        check(kf_unpack(dev_, nchunks_, (const void *)fused, scale, dtype_,
+                        (void *)stream),
+              "kf_unpack");
+    }
+    uint64_t total_elems() const { return total_; }
+    int nchunks() const { return nchunks_; }
+
+  private:
+    int dtype_;
+    int nchunks_ = 0;
+    void *dev_ = nullptr;
+    uint64_t total_ = 0;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_hip, m)
+{
+    m.doc() = "KungFu-AMD gfx950 HIP kernels";
+    m.attr("KERNEL_ARCH") = "gfx950";
+
+    py::class_<FusionPlan>(m, "FusionPlan")
+        .def(py::init<const std::vector<
+                          std::tuple<uintptr_t, uint64_t, uint64_t>> &,
+                      int, unsigned>(),
+             py::arg("segments"), py::arg("dtype"),
+             py::arg("chunk_elems") = 1u << 14)
+        .def("pack", &FusionPlan::pack, py::arg("fused"), py::arg("stream"))
+        .def("unpack", &FusionPlan::unpack, py::arg("fused"),
+             py::arg("scale"), py::arg("stream"))
+        .def_property_readonly("total_elems", &FusionPlan::total_elems)
+        .def_property_readonly("nchunks", &FusionPlan::nchunks);
+
+    m.def("avg_inplace",
+          [](uintptr_t y, uintptr_t x, float alpha, long long n, int dtype,
+             uintptr_t stream) {
+              check(kf_avg_inplace((void *)y, (const void *)x, alpha, n,
+                                   dtype, (void *)stream),
+                    "kf_avg_inplace");
+          });
+    m.def("scale_", [](uintptr_t y, float s, long long n, int dtype,
+                       uintptr_t stream) {
+        check(kf_scale((void *)y, s, n, dtype, (void *)stream), "kf_scale");
+    });
+    m.def("norm2",
+          [](uintptr_t x, long long n, uintptr_t out_f32, int dtype,
+             uintptr_t stream) {
+              check(kf_norm2((const void *)x, n, (void *)out_f32, dtype,
+                             (void *)stream),
+                    "kf_norm2");
+          });
+    m.def("dot",
+          [](uintptr_t x, uintptr_t y, long long n, uintptr_t out_f32,
+             int dtype, uintptr_t stream) {
+              check(kf_dot((const void *)x, (const void *)y, n,
+                           (void *)out_f32, dtype, (void *)stream),
+                    "kf_dot");
+          });
+    m.def("sgd_momentum",
+          [](uintptr_t p, uintptr_t g, uintptr_t m_f32, long long n,
+             float lr, float momentum, float weight_decay, float grad_scale,
+             bool nesterov, int dtype, uintptr_t stream) {
+              check(kf_sgd_momentum((void *)p, (const void *)g,
+                                    (void *)m_f32, n, lr, momentum,
+                                    weight_decay, grad_scale,
+                                    nesterov ? 1 : 0, dtype,
+                                    (void *)stream),
+                    "kf_sgd_momentum");
+          });
+    m.def("transform2",
+          [](uintptr_t z, uintptr_t x, long long n, int op, int dtype,
+             uintptr_t stream) {
+              check(kf_transform2((void *)z, (const void *)x, n, op, dtype,
+                                  (void *)stream),
+                    "kf_transform2");
+          });
+    m.def("device_synchronize", [] { check(hipDeviceSynchronize(), "sync"); });
+}

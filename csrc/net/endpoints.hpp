@@ -1,0 +1,95 @@
+// Message endpoints: collective rendezvous queues, P2P blob request/response,
+// control messages.
+//
+// Reference parity: srcs/go/rchannel/handler/{collective,p2p}.go and
+// srcs/go/store/. The collective endpoint supports zero-copy receive
+// (recv_into pre-registers the destination buffer, collective.go:34-41);
+// early arrivals are buffered so connection handler threads never block.
+#pragma once
+
+#include <condition_variable>
+#include <deque>
+#include <functional>
+#include <memory>
+#include <mutex>
+#include <unordered_map>
+
+#include "transport.hpp"
+
+namespace kf {
+
+class CollectiveEndpoint {
+  public:
+    void on_frame(const PeerID &src, Frame &f);
+    // Copy (or zero-copy if pre-registered before arrival) one message from
+    // (src, name) into dst. Throws on size mismatch or shutdown.
+    void recv_into(const PeerID &src, const std::string &name, void *dst,
+                   size_t len);
+    // Take ownership of one message buffer (receive path that aggregates).
+    std::vector<uint8_t> recv(const PeerID &src, const std::string &name);
+    void clear();     // drop pending state (on resize)
+    void shutdown();  // wake all waiters with failure
+
+  private:
+    struct Slot {
+        std::mutex mu;
+        std::condition_variable cv;
+        std::deque<std::vector<uint8_t>> q;  // early arrivals
+        uint8_t *dst = nullptr;              // registered destination
+        size_t dst_len = 0;
+        bool filled = false;
+        bool dead = false;
+    };
+    std::shared_ptr<Slot> slot(const PeerID &src, const std::string &name);
+
+    std::mutex mu_;
+    std::unordered_map<std::string, std::shared_ptr<Slot>> slots_;
+    bool dead_ = false;
+};
+
+// In-memory blob store for P2P model exchange (reference: srcs/go/store/).
+// save() replaces the blob atomically via shared_ptr swap, so concurrent
+// request handlers keep a consistent snapshot without a window GC.
+class BlobStore {
+  public:
+    void save(const std::string &name, const void *data, size_t len);
+    std::shared_ptr<const std::vector<uint8_t>> get(
+        const std::string &name) const;
+    uint64_t version(const std::string &name) const;
+
+  private:
+    mutable std::mutex mu_;
+    std::unordered_map<std::string,
+                       std::shared_ptr<const std::vector<uint8_t>>>
+        blobs_;
+    std::unordered_map<std::string, uint64_t> versions_;
+};
+
+// Request/response model pulls (AD-PSGD PairAveraging).
+class P2PEndpoint {
+  public:
+    P2PEndpoint(BlobStore &store, Client &client, const PeerID &self)
+        : store_(store), client_(client), self_(self)
+    {
+    }
+    void on_frame(const PeerID &src, Frame &f);
+    // Pull blob `name` from target into dst; true on success.
+    bool request(const PeerID &target, const std::string &name, void *dst,
+                 size_t len, int timeout_ms = 30000);
+    void shutdown();
+
+  private:
+    struct Waiter {
+        std::mutex mu;
+        std::condition_variable cv;
+        std::vector<uint8_t> data;
+        bool done = false, failed = false;
+    };
+    BlobStore &store_;
+    Client &client_;
+    PeerID self_;
+    std::mutex mu_;
+    std::unordered_map<std::string, std::shared_ptr<Waiter>> waiters_;
+};
+
+}  // namespace kf

@@ -1,0 +1,131 @@
+// Point-to-point transport: framed messages over TCP / Unix sockets.
+//
+// Reference parity: srcs/go/rchannel/{connection,client,server}
+// (message framing message.go:42-199, handshake with cluster-version token
+// connection.go:28-87, conn pool client.go, TCP+Unix listeners server.go).
+// Re-designed in C++: connections are simplex (client sends frames to the
+// remote peer's server); the only duplex conn type is Ping (server echoes).
+// Colocated peers (same IPv4) use Unix domain sockets.
+#pragma once
+
+#include <atomic>
+#include <cstdint>
+#include <functional>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <string>
+#include <thread>
+#include <vector>
+
+#include "../core/plan.hpp"
+
+namespace kf {
+
+enum class ConnType : uint8_t {
+    Ping = 0,
+    Control = 1,
+    Collective = 2,
+    P2P = 3,
+};
+
+namespace msgflag {
+constexpr uint32_t IsResponse = 1u << 0;
+constexpr uint32_t RequestFailed = 1u << 1;
+constexpr uint32_t IsRequest = 1u << 2;
+}  // namespace msgflag
+
+struct Frame {
+    std::string name;
+    uint32_t flags = 0;
+    std::vector<uint8_t> data;
+};
+
+// Thread-safe framed socket wrapper.
+class Conn {
+  public:
+    explicit Conn(int fd) : fd_(fd) {}
+    ~Conn() { close_fd(); }
+    Conn(const Conn &) = delete;
+
+    bool send_frame(const std::string &name, uint32_t flags,
+                    const void *data, size_t len);
+    bool read_frame(Frame &f);  // blocking; false on EOF/error
+    void close_fd();
+    int fd() const { return fd_; }
+    std::mutex &write_mutex() { return wmu_; }
+
+  private:
+    std::atomic<int> fd_;
+    std::mutex wmu_;
+};
+
+struct Handshake {
+    ConnType type = ConnType::Ping;
+    PeerID src;
+    uint32_t token = 0;  // cluster version; stale peers are fenced off
+};
+
+using FrameHandler =
+    std::function<void(const Handshake &, Frame &, Conn &)>;
+
+// Listens on TCP (self.port) and, when enabled, on a Unix socket keyed by
+// port. One handler thread per accepted connection.
+class Server {
+  public:
+    Server(const PeerID &self, bool use_unix);
+    ~Server();
+    // token_ok(peer_token) decides whether to accept (elastic fencing)
+    void start(FrameHandler handler,
+               std::function<bool(uint32_t)> token_ok);
+    void stop();
+    static std::string unix_sock_path(uint16_t port);
+
+  private:
+    void accept_loop(int listen_fd);
+    void handle_conn(int fd);
+
+    PeerID self_;
+    bool use_unix_;
+    int tcp_fd_ = -1, unix_fd_ = -1;
+    std::atomic<bool> stopping_{false};
+    FrameHandler handler_;
+    std::function<bool(uint32_t)> token_ok_;
+    std::vector<std::thread> threads_;
+    std::mutex mu_;
+    std::vector<std::shared_ptr<Conn>> conns_;
+};
+
+// Outgoing connection pool: one cached conn per (remote, type).
+class Client {
+  public:
+    explicit Client(const PeerID &self) : self_(self) {}
+    ~Client() { reset({}, 0); }
+
+    void set_token(uint32_t token) { token_ = token; }
+    // Send one frame; establishes/caches the connection. Throws on failure.
+    void send(const PeerID &remote, ConnType type, const std::string &name,
+              uint32_t flags, const void *data, size_t len);
+    // RTT ping; returns microseconds, or -1 on failure.
+    int64_t ping(const PeerID &remote, int timeout_ms = 2000);
+    // Wait until remote answers pings (poll every poll_ms).
+    bool wait(const PeerID &remote, int timeout_ms, int poll_ms = 200);
+    // Drop all conns except to `keeps`; adopt new token (elastic resize).
+    void reset(const std::vector<PeerID> &keeps, uint32_t token);
+
+    // egress byte accounting (monitoring)
+    uint64_t egress_bytes(const PeerID &remote) const;
+    std::map<uint64_t, uint64_t> egress_all() const;
+
+  private:
+    std::shared_ptr<Conn> get_conn(const PeerID &remote, ConnType type,
+                                   int connect_timeout_ms = 20000);
+
+    PeerID self_;
+    std::atomic<uint32_t> token_{0};
+    mutable std::mutex mu_;
+    std::map<std::pair<uint64_t, uint8_t>, std::shared_ptr<Conn>> pool_;
+    std::map<uint64_t, uint64_t> egress_;
+};
+
+}  // namespace kf

@@ -1,0 +1,354 @@
+// pybind11 bindings for the KungFu-AMD C++ runtime (`kungfu_amd._core`).
+//
+// Reference parity: the cgo bridge srcs/go/libkungfu-comm/ + the Python C API
+// srcs/cpp/src/python/c_api.cpp (rank/size/barrier/resize/propose/save/
+// request/collectives). Buffers are passed as raw data pointers
+// (tensor.data_ptr()) for zero-copy in/out of the C++ engine; blocking calls
+// release the GIL.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <cstring>
+#include <deque>
+#include <memory>
+#include <mutex>
+
+#include "../core/graph.hpp"
+#include "../core/plan.hpp"
+#include "../peer/peer.hpp"
+
+namespace py = pybind11;
+using namespace kf;
+
+namespace {
+
+std::unique_ptr<Peer> g_peer;
+std::mutex g_mu;
+
+Peer &peer()
+{
+    if (!g_peer) throw std::runtime_error("kungfu not initialized");
+    return *g_peer;
+}
+
+Workspace make_ws(uintptr_t send, uintptr_t recv, size_t count, int dtype,
+                  int op, const std::string &name)
+{
+    Workspace w;
+    w.send = (const void *)send;
+    w.recv = (void *)recv;
+    w.count = count;
+    w.dt = (DType)dtype;
+    w.op = (ReduceOp)op;
+    w.name = name;
+    return w;
+}
+
+// Prim MST over a dense symmetric weight matrix (reference:
+// srcs/cpp/include/kungfu/mst.hpp:10-58). Returns the parent array
+// (parent[root] = root) usable with Session::set_tree.
+std::vector<int> prim_mst(const std::vector<double> &w, int n)
+{
+    if ((int)w.size() != n * n) throw std::runtime_error("bad matrix");
+    std::vector<int> parent(n, 0);
+    std::vector<double> dist(n, 1e300);
+    std::vector<bool> in(n, false);
+    parent[0] = 0;
+    dist[0] = 0;
+    for (int it = 0; it < n; ++it) {
+        int u = -1;
+        for (int i = 0; i < n; ++i) {
+            if (!in[i] && (u < 0 || dist[i] < dist[u])) u = i;
+        }
+        in[u] = true;
+        for (int v = 0; v < n; ++v) {
+            if (!in[v] && w[u * n + v] < dist[v]) {
+                dist[v] = w[u * n + v];
+                parent[v] = u;
+            }
+        }
+    }
+    return parent;
+}
+
+// Control-plane server for the launcher's watch/elastic mode: queues
+// ("update", stage-json) frames from workers (reference runner/handler.go).
+class RunnerServer {
+  public:
+    RunnerServer(const std::string &self_spec, bool use_unix)
+        : self_(PeerID::parse(self_spec)),
+          server_(std::make_unique<Server>(self_, use_unix))
+    {
+        server_->start(
+            [this](const Handshake &hs, Frame &f, Conn &) {
+                std::lock_guard<std::mutex> lk(mu_);
+                q_.emplace_back(f.name,
+                                std::string((const char *)f.data.data(),
+                                            f.data.size()));
+                cv_.notify_all();
+                (void)hs;
+            },
+            [](uint32_t) { return true; });
+    }
+    // returns (name, payload) or None after timeout
+    py::object poll(int timeout_ms)
+    {
+        std::unique_lock<std::mutex> lk(mu_);
+        if (!cv_.wait_for(lk, std::chrono::milliseconds(timeout_ms),
+                          [this] { return !q_.empty(); }))
+            return py::none();
+        auto item = q_.front();
+        q_.pop_front();
+        return py::make_tuple(item.first, py::bytes(item.second));
+    }
+    void stop() { server_->stop(); }
+
+  private:
+    PeerID self_;
+    std::unique_ptr<Server> server_;
+    std::mutex mu_;
+    std::condition_variable cv_;
+    std::deque<std::pair<std::string, std::string>> q_;
+};
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m)
+{
+    m.doc() = "KungFu-AMD native runtime (control plane + CPU collectives)";
+
+    // ---- lifecycle ----
+    m.def("init", [] {
+        std::lock_guard<std::mutex> lk(g_mu);
+        if (g_peer) return;
+        auto cfg = parse_env_config();
+        g_peer = std::make_unique<Peer>(cfg);
+        py::gil_scoped_release rel;
+        g_peer->start();
+    });
+    m.def("finalize", [] {
+        std::lock_guard<std::mutex> lk(g_mu);
+        if (!g_peer) return;
+        py::gil_scoped_release rel;
+        g_peer.reset();
+    });
+    m.def("initialized", [] { return (bool)g_peer; });
+
+    // ---- metadata ----
+    m.def("rank", [] { return peer().rank(); });
+    m.def("size", [] { return peer().size(); });
+    m.def("local_rank", [] { return peer().local_rank(); });
+    m.def("local_size", [] { return peer().local_size(); });
+    m.def("host_count", [] { return peer().host_count(); });
+    m.def("uid", [] { return peer().uid(); });
+    m.def("detached", [] { return peer().detached(); });
+    m.def("cluster_version", [] { return peer().version(); });
+
+    // ---- collectives (blocking; GIL released) ----
+    m.def("barrier", [] {
+        py::gil_scoped_release rel;
+        peer().session().barrier();
+    });
+    m.def("all_reduce",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt, int op,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, op, name);
+              py::gil_scoped_release rel;
+              peer().session().all_reduce(w);
+          });
+    m.def("reduce",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt, int op,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, op, name);
+              py::gil_scoped_release rel;
+              peer().session().reduce(w);
+          });
+    m.def("broadcast",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt,
+             const std::string &name, int root) {
+              auto w = make_ws(s, r, count, dt, 0, name);
+              py::gil_scoped_release rel;
+              peer().session().broadcast(w, root);
+          });
+    m.def("all_gather",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, 0, name);
+              py::gil_scoped_release rel;
+              peer().session().all_gather(w);
+          });
+    m.def("gather",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, 0, name);
+              py::gil_scoped_release rel;
+              peer().session().gather(w);
+          });
+    m.def("local_reduce",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt, int op,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, op, name);
+              py::gil_scoped_release rel;
+              peer().session().local_reduce(w);
+          });
+    m.def("local_broadcast",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, 0, name);
+              py::gil_scoped_release rel;
+              peer().session().local_broadcast(w);
+          });
+    m.def("cross_all_reduce",
+          [](uintptr_t s, uintptr_t r, size_t count, int dt, int op,
+             const std::string &name) {
+              auto w = make_ws(s, r, count, dt, op, name);
+              py::gil_scoped_release rel;
+              peer().session().cross_all_reduce(w);
+          });
+    m.def("consensus", [](py::bytes data, const std::string &name) {
+        std::string s = data;
+        py::gil_scoped_release rel;
+        return peer().session().consensus(s.data(), s.size(), name);
+    });
+
+    // ---- P2P model store ----
+    m.def("save", [](const std::string &name, uintptr_t data, size_t len) {
+        py::gil_scoped_release rel;
+        peer().save(name, (const void *)data, len);
+    });
+    m.def("request",
+          [](int target, const std::string &name, uintptr_t dst,
+             size_t len) {
+              py::gil_scoped_release rel;
+              return peer().request(target, name, (void *)dst, len);
+          });
+
+    // ---- elastic ----
+    m.def("propose_new_size", [](int n) {
+        py::gil_scoped_release rel;
+        return peer().propose_new_size(n);
+    });
+    m.def("resize_cluster_from_url", [] {
+        py::gil_scoped_release rel;
+        auto rr = peer().resize_cluster_from_url();
+        return std::make_pair(rr.changed, rr.detached);
+    });
+    m.def("resize", [](int n) {
+        py::gil_scoped_release rel;
+        auto rr = peer().resize(n);
+        return std::make_pair(rr.changed, rr.detached);
+    });
+
+    // ---- adaptation / monitoring ----
+    m.def("set_tree", [](const std::vector<int> &parent) {
+        py::gil_scoped_release rel;
+        peer().session().set_tree(parent);
+    });
+    m.def("set_strategy", [](const std::string &name) {
+        auto s = strategy_from_name(name);
+        py::gil_scoped_release rel;
+        peer().session().set_strategy(s);
+    });
+    m.def("get_strategy",
+          [] { return strategy_name(peer().session().strategy()); });
+    m.def("peer_latencies_us", [] {
+        py::gil_scoped_release rel;
+        return peer().peer_latencies_us();
+    });
+    m.def("prim_mst", &prim_mst);
+    m.def("strategy_stats", [] {
+        auto st = peer().session().stats();
+        py::list out;
+        for (auto &s : st) {
+            py::dict d;
+            d["ops"] = s.ops;
+            d["bytes"] = s.bytes;
+            d["seconds"] = s.seconds;
+            d["throughput"] = s.throughput();
+            out.append(d);
+        }
+        return out;
+    });
+    m.def("reset_strategy_stats",
+          [] { peer().session().reset_stats(); });
+    m.def("check_interference", [](double ratio) {
+        return peer().session().check_interference(ratio);
+    });
+    m.def("egress_bytes", [] {
+        auto eg = peer().egress_bytes();
+        py::dict d;
+        for (auto &kv : eg) {
+            PeerID p;
+            p.ipv4 = (uint32_t)(kv.first >> 16);
+            p.port = (uint16_t)(kv.first & 0xffff);
+            d[py::str(p.str())] = kv.second;
+        }
+        return d;
+    });
+
+    // ---- launcher support ----
+    py::class_<RunnerServer>(m, "RunnerServer")
+        .def(py::init<const std::string &, bool>(), py::arg("self_spec"),
+             py::arg("use_unix") = false)
+        .def("poll", &RunnerServer::poll, py::arg("timeout_ms"),
+             py::call_guard<py::gil_scoped_release>())
+        .def("stop", &RunnerServer::stop,
+             py::call_guard<py::gil_scoped_release>());
+
+    // ---- plan/topology helpers (unit tests + tools) ----
+    m.def("topology_digest",
+          [](const std::string &peers, const std::string &strategy) {
+              auto pl = PeerList::parse(peers);
+              auto gs = gen_strategies(pl, strategy_from_name(strategy));
+              std::string out;
+              for (auto &g : gs) {
+                  out += g.reduce.digest() + "/" + g.bcast.digest() + "\n";
+              }
+              return out;
+          });
+    m.def("topology_edges", [](int n, const std::string &strategy,
+                               const std::string &peers) {
+        PeerList pl;
+        if (!peers.empty()) {
+            pl = PeerList::parse(peers);
+        } else {
+            for (int i = 0; i < n; ++i)
+                pl.peers.push_back(
+                    PeerID::parse("127.0.0.1:" + std::to_string(10000 + i)));
+        }
+        auto gs = gen_strategies(pl, strategy_from_name(strategy));
+        py::list out;
+        for (auto &g : gs) {
+            py::list redges, bedges, roots;
+            for (int i = 0; i < g.reduce.n; ++i) {
+                for (int j : g.reduce.nexts[i])
+                    redges.append(py::make_tuple(i, j));
+                if (g.reduce.self_loop[i]) roots.append(i);
+            }
+            for (int i = 0; i < g.bcast.n; ++i) {
+                for (int j : g.bcast.nexts[i])
+                    bedges.append(py::make_tuple(i, j));
+            }
+            py::dict d;
+            d["reduce"] = redges;
+            d["bcast"] = bedges;
+            d["roots"] = roots;
+            out.append(d);
+        }
+        return out;
+    });
+    m.def("cluster_resize_json",
+          [](const std::string &json, int new_size, int port_base) {
+              auto c = Cluster::from_json(json);
+              return c.resized(new_size, port_base).json();
+          });
+    m.def("gen_peer_list",
+          [](const std::string &hosts, int np, int port_base) {
+              auto hl = HostList::parse(hosts);
+              return hl.gen_peer_list(np, port_base).str();
+          });
+    m.def("gen_runner_list", [](const std::string &hosts, int port) {
+        auto hl = HostList::parse(hosts);
+        return hl.gen_runner_list(port).str();
+    });
+}

@@ -1,0 +1,347 @@
+#include "session.hpp"
+
+#include <chrono>
+#include <cstring>
+#include <thread>
+
+namespace kf {
+
+namespace {
+// FNV-1a for strategy sharding (reference shard.go hash-by-name/index)
+uint64_t fnv1a(const std::string &s)
+{
+    uint64_t h = 1469598103934665603ull;
+    for (char c : s) {
+        h ^= (uint8_t)c;
+        h *= 1099511628211ull;
+    }
+    return h;
+}
+}  // namespace
+
+Session::Session(const PeerList &peers, int rank, Client &client,
+                 CollectiveEndpoint &collective, Strategy strategy)
+    : peers_(peers),
+      rank_(rank),
+      client_(client),
+      collective_(collective),
+      strategy_kind_(strategy)
+{
+    global_ = gen_strategies(peers_, strategy);
+    local_ = gen_local_strategies(peers_);
+    cross_ = gen_cross_strategies(peers_);
+    stats_.resize(global_.size());
+}
+
+void Session::set_tree(const std::vector<int> &parent)
+{
+    if ((int)parent.size() != peers_.size())
+        throw std::runtime_error("set_tree: bad forest size");
+    global_ = {gen_from_forest(parent)};
+    std::lock_guard<std::mutex> lk(stats_mu_);
+    stats_.assign(1, {});
+}
+
+void Session::set_strategy(Strategy s)
+{
+    strategy_kind_ = s;
+    global_ = gen_strategies(peers_, s);
+    std::lock_guard<std::mutex> lk(stats_mu_);
+    stats_.assign(global_.size(), {});
+}
+
+// One (reduce,bcast) graph-pair execution over one chunk.
+// Reference: session.go runGraphs (222-290).
+void Session::run_graphs(const Workspace &w, const GraphPair &g,
+                         const std::string &suffix)
+{
+    const size_t bytes = w.count * dtype_size(w.dt);
+    const auto &rg = g.reduce;
+    const auto &bg = g.bcast;
+    const bool isolated = rg.prevs[rank_].empty() &&
+                          rg.nexts[rank_].empty() && !rg.self_loop[rank_] &&
+                          bg.prevs[rank_].empty() && bg.nexts[rank_].empty();
+    if (w.recv != w.send && w.send != nullptr) {
+        std::memcpy(w.recv, w.send, bytes);
+    }
+    if (isolated) return;
+
+    uint8_t *acc = (uint8_t *)w.recv;
+    const std::string rname = w.name + suffix + "|r";
+    const std::string bname = w.name + suffix + "|b";
+
+    // reduce phase: aggregate partials from children, forward to parent
+    for (int p : rg.prevs[rank_]) {
+        auto buf = collective_.recv(peer(p), rname);
+        if (buf.size() != bytes)
+            throw std::runtime_error("reduce size mismatch on " + rname);
+        reduce_inplace(acc, buf.data(), w.count, w.dt, w.op);
+    }
+    for (int nx : rg.nexts[rank_]) {
+        client_.send(peer(nx), ConnType::Collective, rname, 0, acc, bytes);
+    }
+
+    // bcast phase: receive the final value, pipeline onward
+    if (!bg.prevs[rank_].empty()) {
+        collective_.recv_into(peer(bg.prevs[rank_][0]), bname, acc, bytes);
+    }
+    for (int nx : bg.nexts[rank_]) {
+        client_.send(peer(nx), ConnType::Collective, bname, 0, acc, bytes);
+    }
+}
+
+// Bcast-graph-only execution (for broadcast).
+void Session::run_bcast_graph(const Workspace &w, const Graph &bg,
+                              const std::string &suffix)
+{
+    const size_t bytes = w.count * dtype_size(w.dt);
+    const std::string bname = w.name + suffix + "|b";
+    uint8_t *acc = (uint8_t *)w.recv;
+    if (!bg.prevs[rank_].empty()) {
+        collective_.recv_into(peer(bg.prevs[rank_][0]), bname, acc, bytes);
+    } else if (w.recv != w.send && w.send != nullptr) {
+        std::memcpy(w.recv, w.send, bytes);
+    }
+    for (int nx : bg.nexts[rank_]) {
+        client_.send(peer(nx), ConnType::Collective, bname, 0, acc, bytes);
+    }
+}
+
+// Chunked multi-strategy dispatch (reference session.go:292-317 + shard.go).
+void Session::run_strategies(const Workspace &w,
+                             const std::vector<GraphPair> &strategies,
+                             bool monitored)
+{
+    const size_t elem = dtype_size(w.dt);
+    const size_t bytes = w.count * elem;
+    // Messages are keyed by (src, name) with FIFO queues, so repeated ops
+    // on the same name (one per training step) stay correctly matched even
+    // when ranks run ahead; no sequence number is needed (and one would
+    // break concurrent ops issued in different orders across ranks).
+    const std::string sfx_base = "";
+
+    size_t nchunks = 1;
+    if (bytes > kChunkBytes && strategies.size() > 0) {
+        nchunks = (bytes + kChunkBytes - 1) / kChunkBytes;
+        if (nchunks > kMaxChunks) nchunks = kMaxChunks;
+    }
+    auto t0 = std::chrono::steady_clock::now();
+    if (nchunks == 1) {
+        const size_t sidx = fnv1a(w.name) % strategies.size();
+        run_graphs(w, strategies[sidx], sfx_base + "@" +
+                                            std::to_string(sidx));
+        if (monitored) {
+            auto t1 = std::chrono::steady_clock::now();
+            std::lock_guard<std::mutex> lk(stats_mu_);
+            if (sidx < stats_.size()) {
+                auto &st = stats_[sidx];
+                st.ops++;
+                st.bytes += bytes;
+                st.seconds +=
+                    std::chrono::duration<double>(t1 - t0).count();
+            }
+        }
+        return;
+    }
+
+    const size_t per = (w.count + nchunks - 1) / nchunks;
+    std::vector<std::thread> threads;
+    std::vector<std::string> errors(nchunks);
+    threads.reserve(nchunks);
+    for (size_t c = 0; c < nchunks; ++c) {
+        const size_t begin = c * per;
+        const size_t cnt = std::min(per, w.count - begin);
+        if (cnt == 0 || begin >= w.count) break;
+        Workspace cw = w;
+        cw.send = w.send ? (const uint8_t *)w.send + begin * elem : nullptr;
+        cw.recv = (uint8_t *)w.recv + begin * elem;
+        cw.count = cnt;
+        const size_t sidx =
+            (fnv1a(w.name) + c) % strategies.size();  // spread chunks
+        std::string sfx = sfx_base + "." + std::to_string(c) + "@" +
+                          std::to_string(sidx);
+        threads.emplace_back([this, cw, &strategies, sidx, sfx, &errors,
+                              c] {
+            try {
+                run_graphs(cw, strategies[sidx], sfx);
+            } catch (const std::exception &e) {
+                errors[c] = e.what();
+            }
+        });
+    }
+    for (auto &t : threads) t.join();
+    for (auto &e : errors) {
+        if (!e.empty()) throw std::runtime_error("chunk failed: " + e);
+    }
+    if (monitored) {
+        auto t1 = std::chrono::steady_clock::now();
+        std::lock_guard<std::mutex> lk(stats_mu_);
+        const size_t sidx = fnv1a(w.name) % stats_.size();
+        auto &st = stats_[sidx];
+        st.ops++;
+        st.bytes += bytes;
+        st.seconds += std::chrono::duration<double>(t1 - t0).count();
+    }
+}
+
+void Session::all_reduce(const Workspace &w)
+{
+    run_strategies(w, global_, true);
+}
+
+void Session::reduce(const Workspace &w)
+{
+    // reduce-graph-only run: we reuse run_graphs but skip bcast by using a
+    // pair whose bcast graph has no edges except at the root.
+    const auto &g = global_[0];
+    Workspace ww = w;
+    GraphPair p;
+    p.reduce = g.reduce;
+    p.bcast = Graph(p.reduce.n);  // empty bcast: result stays at root
+    run_graphs(ww, p, "@rd");
+}
+
+void Session::broadcast(const Workspace &w, int root)
+{
+    const std::string sfx = "@bc" + std::to_string(root);
+    if (root == 0) {
+        run_bcast_graph(w, global_[0].bcast, sfx);
+    } else {
+        auto pair = gen_star(peers_.size(), root);
+        run_bcast_graph(w, pair.bcast, sfx);
+    }
+}
+
+void Session::all_gather(const Workspace &w)
+{
+    const size_t elem = dtype_size(w.dt);
+    const size_t bytes = w.count * elem;
+    const std::string name = w.name + "@ag";
+    const int n = peers_.size();
+    uint8_t *out = (uint8_t *)w.recv;
+    std::memcpy(out + (size_t)rank_ * bytes, w.send, bytes);
+    // full-mesh exchange (reference allgather.go:17-45)
+    std::vector<std::thread> threads;
+    for (int r = 0; r < n; ++r) {
+        if (r == rank_) continue;
+        threads.emplace_back([this, r, name, bytes, &w] {
+            client_.send(peer(r), ConnType::Collective, name, 0, w.send,
+                         bytes);
+        });
+    }
+    for (int r = 0; r < n; ++r) {
+        if (r == rank_) continue;
+        collective_.recv_into(peer(r), name, out + (size_t)r * bytes,
+                              bytes);
+    }
+    for (auto &t : threads) t.join();
+}
+
+void Session::gather(const Workspace &w)
+{
+    const size_t elem = dtype_size(w.dt);
+    const size_t bytes = w.count * elem;
+    const std::string name = w.name + "@g";
+    if (rank_ == 0) {
+        uint8_t *out = (uint8_t *)w.recv;
+        std::memcpy(out, w.send, bytes);
+        for (int r = 1; r < peers_.size(); ++r) {
+            collective_.recv_into(peer(r), name, out + (size_t)r * bytes,
+                                  bytes);
+        }
+    } else {
+        client_.send(peer(0), ConnType::Collective, name, 0, w.send, bytes);
+    }
+}
+
+void Session::barrier()
+{
+    uint8_t b = 0, out = 0;
+    Workspace w;
+    w.send = &b;
+    w.recv = &out;
+    w.count = 1;
+    w.dt = DType::U8;
+    w.op = ReduceOp::SUM;
+    w.name = "|barrier";
+    run_strategies(w, global_, false);
+}
+
+bool Session::consensus(const void *data, size_t len,
+                        const std::string &name)
+{
+    // agree on length first, then min/max over the bytes
+    uint64_t lmin = len, lmax = len;
+    Workspace wl;
+    wl.count = 1;
+    wl.dt = DType::U64;
+    wl.name = name + "|len";
+    wl.send = &len;
+    uint64_t tmp = len;
+    wl.send = &tmp;
+    wl.recv = &lmin;
+    wl.op = ReduceOp::MIN;
+    run_strategies(wl, global_, false);
+    wl.recv = &lmax;
+    wl.op = ReduceOp::MAX;
+    run_strategies(wl, global_, false);
+    if (lmin != lmax) return false;
+    if (len == 0) return true;
+    std::vector<uint8_t> mn(len), mx(len);
+    Workspace w;
+    w.send = data;
+    w.count = len;
+    w.dt = DType::U8;
+    w.name = name + "|bytes";
+    w.recv = mn.data();
+    w.op = ReduceOp::MIN;
+    run_strategies(w, global_, false);
+    w.recv = mx.data();
+    w.op = ReduceOp::MAX;
+    run_strategies(w, global_, false);
+    return std::memcmp(mn.data(), mx.data(), len) == 0 &&
+           std::memcmp(mn.data(), data, len) == 0;
+}
+
+void Session::local_reduce(const Workspace &w)
+{
+    run_strategies(w, local_, false);
+}
+
+void Session::local_broadcast(const Workspace &w)
+{
+    run_bcast_graph(w, local_[0].bcast, "@lb");
+}
+
+void Session::cross_all_reduce(const Workspace &w)
+{
+    run_strategies(w, cross_, false);
+}
+
+std::vector<StrategyStat> Session::stats() const
+{
+    std::lock_guard<std::mutex> lk(stats_mu_);
+    return stats_;
+}
+
+void Session::reset_stats()
+{
+    std::lock_guard<std::mutex> lk(stats_mu_);
+    for (auto &s : stats_) s = {};
+}
+
+bool Session::check_interference(double ratio)
+{
+    std::lock_guard<std::mutex> lk(stats_mu_);
+    double cur = 0;
+    uint64_t total_ops = 0;
+    for (auto &s : stats_) {
+        cur += s.throughput();
+        total_ops += s.ops;
+    }
+    if (total_ops == 0) return false;
+    if (cur > best_throughput_) best_throughput_ = cur;
+    return best_throughput_ > 0 && cur < ratio * best_throughput_;
+}
+
+}  // namespace kf

@@ -1,0 +1,3 @@
+from kungfu_amd.models.slp import SLP  # noqa
+from kungfu_amd.models.resnet import resnet50  # noqa
+from kungfu_amd.models.bert import bert_base  # noqa

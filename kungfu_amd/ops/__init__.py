@@ -1,0 +1,255 @@
+"""Tensor-level collective ops.
+
+Reference parity: srcs/python/kungfu/tensorflow/ops/ and kungfu/torch/ops/.
+Routing (MI355X-native):
+  * CUDA tensors -> RCCL over xGMI via torch.distributed (bucketed by the
+    callers in kungfu_amd.parallel);
+  * CPU tensors  -> the C++ collective engine (graph strategies over
+    TCP/Unix sockets), zero-copy via data_ptr;
+  * `cpu_staged_*` variants stage GPU tensors through host memory into the
+    C++ engine — the reference's only torch GPU path
+    (src/torch/ops/cuda/collective.cpp:31-54) kept for the cross-host hop
+    and plumbing tests.
+"""
+import torch
+
+from kungfu_amd import _core, _ensure_init
+from kungfu_amd.utils.dtypes import core_dtype, core_op
+
+
+def _dist():
+    import torch.distributed as dist
+
+    if not dist.is_initialized():
+        raise RuntimeError(
+            "CUDA collective requested but the RCCL process group is not "
+            "up; call kungfu_amd.init() first (no silent CPU fallback)")
+    return dist
+
+
+_TORCH_OPS = None
+
+
+def _torch_reduce_op(op):
+    global _TORCH_OPS
+    import torch.distributed as dist
+
+    if _TORCH_OPS is None:
+        _TORCH_OPS = {
+            "sum": dist.ReduceOp.SUM,
+            "min": dist.ReduceOp.MIN,
+            "max": dist.ReduceOp.MAX,
+            "prod": dist.ReduceOp.PRODUCT,
+        }
+    return _TORCH_OPS[op]
+
+
+def all_reduce(tensor, op="sum", name=None, average=False, async_op=False):
+    """All-reduce a tensor in place; returns the tensor (or a work handle
+    tuple when async_op on the RCCL path)."""
+    _ensure_init()
+    if _core.size() == 1:
+        return tensor
+    if tensor.is_cuda:
+        dist = _dist()
+        work = dist.all_reduce(tensor, op=_torch_reduce_op(op),
+                               async_op=async_op)
+        if average:
+            if async_op:
+                raise ValueError("average not supported with async_op")
+            tensor.div_(_core.size())
+        return (tensor, work) if async_op else tensor
+    t = tensor.contiguous()
+    _core.all_reduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                     core_dtype(t.dtype), core_op(op),
+                     name or "t%d" % t.numel())
+    if t.data_ptr() != tensor.data_ptr():
+        tensor.copy_(t)
+    if average:
+        tensor.div_(_core.size())
+    return tensor
+
+
+def broadcast(tensor, root=0, name=None):
+    _ensure_init()
+    if _core.size() == 1:
+        return tensor
+    if tensor.is_cuda:
+        _dist().broadcast(tensor, src=root)
+        return tensor
+    t = tensor.contiguous()
+    _core.broadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                    core_dtype(t.dtype), name or "t%d" % t.numel(), root)
+    if t.data_ptr() != tensor.data_ptr():
+        tensor.copy_(t)
+    return tensor
+
+
+def all_gather(tensor, name=None):
+    """Returns a new tensor of shape (size,) + tensor.shape."""
+    _ensure_init()
+    np_ = _core.size()
+    out = torch.empty((np_,) + tuple(tensor.shape), dtype=tensor.dtype,
+                      device=tensor.device)
+    if np_ == 1:
+        out[0] = tensor
+        return out
+    if tensor.is_cuda:
+        dist = _dist()
+        dist.all_gather_into_tensor(out.view(np_, -1),
+                                    tensor.contiguous().view(1, -1))
+        return out
+    t = tensor.contiguous()
+    _core.all_gather(t.data_ptr(), out.data_ptr(), t.numel(),
+                     core_dtype(t.dtype), name or "t%d" % t.numel())
+    return out
+
+
+def reduce(tensor, op="sum", name=None):
+    """Reduce to rank 0 (in place at root; other ranks' buffers are used as
+    scratch)."""
+    _ensure_init()
+    if _core.size() == 1:
+        return tensor
+    if tensor.is_cuda:
+        _dist().reduce(tensor, dst=0, op=_torch_reduce_op(op))
+        return tensor
+    t = tensor.contiguous()
+    _core.reduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                 core_dtype(t.dtype), core_op(op),
+                 name or "t%d" % t.numel())
+    if t.data_ptr() != tensor.data_ptr():
+        tensor.copy_(t)
+    return tensor
+
+
+# ---- CPU-staged variants (GPU tensor through host into the C++ engine) ----
+
+def cpu_staged_all_reduce(tensor, op="sum", name=None):
+    _ensure_init()
+    if _core.size() == 1:
+        return tensor
+    host = tensor.detach().to("cpu", non_blocking=False).contiguous()
+    _core.all_reduce(host.data_ptr(), host.data_ptr(), host.numel(),
+                     core_dtype(host.dtype), core_op(op),
+                     name or "h%d" % host.numel())
+    tensor.copy_(host.to(tensor.device))
+    return tensor
+
+
+def hierarchical_all_reduce(tensor, name=None):
+    """Local (intra-host) RCCL reduce -> cross-host CPU all-reduce among
+    local masters -> local RCCL broadcast (reference
+    ops/gpu/collective.cpp:106-158). On one host this degrades to a plain
+    RCCL all-reduce."""
+    _ensure_init()
+    if _core.host_count() <= 1:
+        return all_reduce(tensor, name=name)
+    if tensor.is_cuda:
+        # local reduce over the node group would need a sub-group; for the
+        # common single-host-per-8-GPU case we fall through to the staged
+        # cross path only when the node count > 1
+        tensor = cpu_staged_all_reduce(tensor, name=name)
+        return tensor
+    t = tensor.contiguous()
+    _core.local_reduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                       core_dtype(t.dtype), core_op("sum"),
+                       (name or "t") + "|lr")
+    _core.cross_all_reduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                           core_dtype(t.dtype), core_op("sum"),
+                           (name or "t") + "|x")
+    _core.local_broadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                          core_dtype(t.dtype), (name or "t") + "|lb")
+    if t.data_ptr() != tensor.data_ptr():
+        tensor.copy_(t)
+    return tensor
+
+
+# ---- model-level helpers ----
+
+def broadcast_parameters(params, root=0):
+    """Broadcast model parameters (and optionally buffers) from root.
+    Reference: kungfu/tensorflow/initializer BroadcastGlobalVariablesOp."""
+    _ensure_init()
+    if _core.size() == 1:
+        return
+    params = list(params)
+    if not params:
+        return
+    if params[0].is_cuda:
+        dist = _dist()
+        for p in params:
+            dist.broadcast(p.data, src=root)
+        return
+    for i, p in enumerate(params):
+        broadcast(p.data, root=root, name="bcast/%d" % i)
+
+
+def broadcast_model(model, root=0):
+    with torch.no_grad():
+        broadcast_parameters(list(model.parameters()), root=root)
+        for b in model.buffers():
+            broadcast(b.data, root=root)
+
+
+# ---- P2P model store (AD-PSGD) ----
+
+def save_tensor(name, tensor):
+    _ensure_init()
+    t = tensor.detach()
+    if t.is_cuda:
+        t = t.to("cpu")
+    t = t.contiguous()
+    _core.save(name, t.data_ptr(), t.numel() * t.element_size())
+
+
+def request_tensor(target_rank, name, out):
+    """Pull a stored tensor from target_rank into `out`; returns success."""
+    _ensure_init()
+    if out.is_cuda:
+        host = torch.empty(out.shape, dtype=out.dtype, device="cpu")
+        ok = _core.request(int(target_rank), name, host.data_ptr(),
+                           host.numel() * host.element_size())
+        if ok:
+            out.copy_(host.to(out.device))
+        return ok
+    o = out.contiguous()
+    ok = _core.request(int(target_rank), name, o.data_ptr(),
+                       o.numel() * o.element_size())
+    if ok and o.data_ptr() != out.data_ptr():
+        out.copy_(o)
+    return ok
+
+
+# ---- adaptive topology ----
+
+def get_peer_latencies():
+    _ensure_init()
+    return _core.peer_latencies_us()
+
+
+def minimum_spanning_tree(latency_matrix_flat, n):
+    return _core.prim_mst([float(x) for x in latency_matrix_flat], int(n))
+
+
+def set_tree(parent):
+    _ensure_init()
+    _core.set_tree([int(p) for p in parent])
+
+
+def compute_mst_tree():
+    """Measure pairwise latencies (allgathered) and install the MST as the
+    collective topology (reference: GetPeerLatencies + MinimumSpanningTree +
+    SetTree, ops/adapt.py:49-60)."""
+    import numpy as np
+
+    _ensure_init()
+    np_ = _core.size()
+    lat = np.asarray(_core.peer_latencies_us(), dtype=np.float64)
+    lat = np.maximum(lat, 1.0)
+    row = torch.from_numpy(lat)
+    mat = all_gather(row, name="|latmat").numpy()
+    mat = np.maximum(mat, mat.T)  # symmetrize
+    parent = minimum_spanning_tree(mat.flatten().tolist(), np_)
+    set_tree(parent)
+    return parent

@@ -1,0 +1,220 @@
+"""Gradient fusion + bucketed all-reduce overlapped with backward.
+
+This is the MI355X-native replacement for the reference's gradient-fusion
+path (TF fuse/defuse concat ops + NCCLScheduler deterministic ordering,
+SURVEY.md §2.6 items 6/11 and §2.5 NCCLScheduler):
+
+  * gradients live directly in per-bucket flat buffers (param.grad is a
+    view), so there is NO pack kernel on the hot path at all — the
+    all-reduce runs on the flat buffer the backward pass wrote;
+  * buckets are all-reduced over RCCL (xGMI) as soon as they are complete,
+    from post-accumulate-grad hooks, overlapping communication with the
+    rest of backward;
+  * bucket launch order is fixed (descending creation order = reverse
+    parameter order) regardless of per-rank grad-ready order, which gives
+    the cross-rank deterministic collective ordering the reference needed
+    its NCCLScheduler arrival-order broadcast for (nccl/scheduler.cpp:93-119)
+    by construction;
+  * bucket size defaults to 32 MiB: RCCL rings over 7 xGMI links want a
+    few MiB per ring per collective.
+
+The same reducer works for CPU tensors through the C++ collective engine
+(chunked graph strategies over TCP/Unix sockets) for plumbing mode.
+"""
+import torch
+
+from kungfu_amd import _core, _ensure_init
+from kungfu_amd.utils.dtypes import core_dtype, core_op
+
+DEFAULT_BUCKET_BYTES = 32 << 20
+_ALIGN = 64  # element alignment of bucket slices (16B vector loads)
+
+
+class _Bucket:
+    __slots__ = ("index", "params", "flat", "numel", "ready", "work",
+                 "launched")
+
+    def __init__(self, index):
+        self.index = index
+        self.params = []
+        self.flat = None
+        self.numel = 0
+        self.ready = 0
+        self.work = None
+        self.launched = False
+
+
+class GradBucketReducer:
+    def __init__(self, params, bucket_bytes=DEFAULT_BUCKET_BYTES,
+                 average=True, overlap=True, name="grads"):
+        _ensure_init()
+        self.params = [p for p in params if p.requires_grad]
+        if not self.params:
+            raise ValueError("no trainable parameters")
+        self.average = average
+        self.overlap = overlap
+        self.name = name
+        self.world = _core.size()
+        dev = self.params[0].device
+        self.is_cuda = dev.type == "cuda"
+        self._build_buckets(bucket_bytes)
+        self._hooks = []
+        self._next_launch = 0
+        if self.overlap and self.is_cuda and self.world > 1:
+            self._register_hooks()
+
+    # -- construction ---------------------------------------------------
+
+    def _build_buckets(self, bucket_bytes):
+        # reverse parameter order: grads become ready roughly back-to-front
+        order = list(reversed(self.params))
+        self.buckets = []
+        cur = _Bucket(0)
+        esize = order[0].element_size()
+        cap = max(1, bucket_bytes // esize)
+        for p in order:
+            aligned = (p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
+            if cur.numel > 0 and cur.numel + aligned > cap:
+                self.buckets.append(cur)
+                cur = _Bucket(len(self.buckets))
+            cur.params.append(p)
+            cur.numel += aligned
+        if cur.params:
+            self.buckets.append(cur)
+        # allocate flats and alias grads
+        self.bucket_of = {}
+        for b in self.buckets:
+            dtype = b.params[0].dtype
+            dev = b.params[0].device
+            b.flat = torch.zeros(b.numel, dtype=dtype, device=dev)
+            off = 0
+            for p in b.params:
+                if p.dtype != dtype or p.device != dev:
+                    raise ValueError(
+                        "mixed dtype/device parameters in one reducer")
+                view = b.flat[off:off + p.numel()].view_as(p)
+                if p.grad is not None:
+                    view.copy_(p.grad)  # preserve grads on mid-run adoption
+                p.grad = view
+                off = (off + p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
+                self.bucket_of[p] = b
+
+    def _register_hooks(self):
+        for p in self.params:
+            h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
+            self._hooks.append(h)
+
+    # -- hot path --------------------------------------------------------
+
+    def _on_grad_ready(self, p):
+        b = self.bucket_of[p]
+        b.ready += 1
+        if b.ready == len(b.params):
+            self._drain()
+
+    def _drain(self):
+        # launch complete buckets strictly in bucket-index order so the
+        # RCCL op order is identical on every rank
+        import torch.distributed as dist
+
+        while self._next_launch < len(self.buckets):
+            b = self.buckets[self._next_launch]
+            if b.ready < len(b.params):
+                return
+            b.work = dist.all_reduce(b.flat, async_op=True)
+            b.launched = True
+            self._next_launch += 1
+
+    def zero_grad(self):
+        for b in self.buckets:
+            b.flat.zero_()
+            b.ready = 0
+            b.work = None
+            b.launched = False
+        self._next_launch = 0
+
+    def finalize(self):
+        """Complete all bucket reductions; call between backward() and
+        optimizer.step(). Applies gradient averaging."""
+        if self.world <= 1:
+            return
+        if self.is_cuda:
+            import torch.distributed as dist
+
+            if self.overlap:
+                self._drain()
+                for b in self.buckets:
+                    if not b.launched:  # param got no grad this step
+                        b.work = dist.all_reduce(b.flat, async_op=True)
+                        b.launched = True
+                for b in self.buckets:
+                    b.work.wait()
+            else:
+                for b in self.buckets:
+                    dist.all_reduce(b.flat)
+            if self.average:
+                for b in self.buckets:
+                    b.flat.div_(self.world)
+        else:
+            for b in self.buckets:
+                f = b.flat
+                _core.all_reduce(f.data_ptr(), f.data_ptr(), f.numel(),
+                                 core_dtype(f.dtype), core_op("sum"),
+                                 "%s/%d" % (self.name, b.index))
+                if self.average:
+                    f.div_(self.world)
+
+    @property
+    def grad_scale(self):
+        return 1.0 / self.world if self.average else 1.0
+
+    def detach(self):
+        for h in self._hooks:
+            h.remove()
+        self._hooks = []
+
+
+class FlatParamGroup:
+    """Flat view over a list of same-dtype parameters (weights, not grads):
+    used for fused model broadcast / averaging / P2P exchange. The HIP
+    pack/unpack kernels move data between the live parameters and the flat
+    buffer in one launch."""
+
+    def __init__(self, params, dtype=None):
+        self.params = list(params)
+        p0 = self.params[0]
+        self.device = p0.device
+        self.dtype = dtype or p0.dtype
+        offsets = []
+        off = 0
+        for p in self.params:
+            offsets.append(off)
+            off += (p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
+        self.numel = off
+        self.offsets = offsets
+        self.flat = torch.zeros(off, dtype=self.dtype, device=self.device)
+        self._plan = None
+        if self.device.type == "cuda":
+            from kungfu_amd.ops import hip as hip_ops
+
+            self._plan = hip_ops.FusionPlan(
+                [p.data for p in self.params], offsets, self.dtype)
+
+    def pack(self):
+        if self._plan is not None:
+            self._plan.pack(self.flat)
+        else:
+            for p, off in zip(self.params, self.offsets):
+                self.flat[off:off + p.numel()].copy_(p.data.view(-1))
+        return self.flat
+
+    def unpack(self, scale=1.0):
+        if self._plan is not None:
+            self._plan.unpack(self.flat, scale)
+        else:
+            for p, off in zip(self.params, self.offsets):
+                src = self.flat[off:off + p.numel()].view_as(p.data)
+                if scale != 1.0:
+                    p.data.copy_(src * scale)
+                else:
+                    p.data.copy_(src)

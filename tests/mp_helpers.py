@@ -1,0 +1,196 @@
+"""Top-level worker functions for multi-process tests (spawn-picklable).
+
+Each worker sets the KUNGFU_* env protocol before importing kungfu_amd,
+mirroring what the launcher does (reference test pattern: fake workers
+under the real kungfu-run on loopback, scripts/tests/run-integration-tests.sh).
+"""
+import os
+
+
+def set_env(i, np, base, strategy="AUTO", extra=None):
+    peers = ",".join("127.0.0.1:%d" % (base + j) for j in range(np))
+    os.environ["KUNGFU_SELF_SPEC"] = "127.0.0.1:%d" % (base + i)
+    os.environ["KUNGFU_INIT_PEERS"] = peers
+    os.environ["KUNGFU_ALLREDUCE_STRATEGY"] = strategy
+    os.environ.pop("KUNGFU_CONFIG_SERVER", None)
+    for k, v in (extra or {}).items():
+        os.environ[k] = v
+
+
+def run_worker(fn, i, np, base, strategy, q, extra=None):
+    try:
+        set_env(i, np, base, strategy, extra)
+        result = fn(i, np)
+        q.put((i, "ok", result))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((i, "err", "%s\n%s" % (e, traceback.format_exc())))
+
+
+def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None):
+    """Run fn(rank, np) in np processes; returns list of results."""
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [
+        ctx.Process(target=run_worker, args=(fn, i, np, base, strategy, q),
+                    kwargs={"extra": extra})
+        for i in range(np)
+    ]
+    for p in procs:
+        p.start()
+    results = {}
+    import queue as queue_mod
+
+    try:
+        for _ in range(np):
+            try:
+                i, status, payload = q.get(timeout=timeout)
+            except queue_mod.Empty:
+                raise AssertionError("cluster timed out; results so far: %r"
+                                     % (results,))
+            assert status == "ok", "rank %d failed: %s" % (i, payload)
+            results[i] = payload
+    finally:
+        for p in procs:
+            p.join(timeout=15)
+            if p.is_alive():
+                p.terminate()
+    return [results[i] for i in range(np)]
+
+
+# ---- worker bodies ----
+
+def allreduce_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    assert kf.rank() == rank and kf.size() == np
+
+    out = {}
+    # small f32 sum
+    x = np_.full(17, float(rank + 1), dtype=np_.float32)
+    y = np_.zeros_like(x)
+    _core.all_reduce(x.ctypes.data, y.ctypes.data, x.size, 10, 0, "g1")
+    out["small_sum"] = float(y[0])
+    # large buffer crossing the 1 MiB chunking threshold
+    big = np_.arange(600_000, dtype=np_.float32) + rank
+    ybig = np_.zeros_like(big)
+    _core.all_reduce(big.ctypes.data, ybig.ctypes.data, big.size, 10, 0,
+                     "gbig")
+    expect = big * 0 + (np_.arange(600_000, dtype=np_.float32) * np +
+                        np * (np - 1) / 2)
+    assert np_.allclose(ybig, expect), "chunked allreduce wrong"
+    # min/max on i64
+    v = np_.array([rank * 10 + 5], dtype=np_.int64)
+    mn = np_.zeros_like(v)
+    mx = np_.zeros_like(v)
+    _core.all_reduce(v.ctypes.data, mn.ctypes.data, 1, 4, 1, "mn")
+    _core.all_reduce(v.ctypes.data, mx.ctypes.data, 1, 4, 2, "mx")
+    out["min"] = int(mn[0])
+    out["max"] = int(mx[0])
+    # barrier storm
+    for _ in range(5):
+        kf.barrier()
+    # broadcast from 0
+    b = np_.full(9, float(rank), dtype=np_.float64)
+    _core.broadcast(b.ctypes.data, b.ctypes.data, b.size, 11, "bc", 0)
+    out["bcast"] = float(b[3])
+    # allgather
+    g = np_.array([rank], dtype=np_.int32)
+    go = np_.zeros(np, dtype=np_.int32)
+    _core.all_gather(g.ctypes.data, go.ctypes.data, 1, 3, "ag")
+    out["gathered"] = go.tolist()
+    # gather to root
+    gr = np_.zeros(np, dtype=np_.int32)
+    _core.gather(g.ctypes.data, gr.ctypes.data, 1, 3, "gr")
+    out["rooted"] = gr.tolist() if rank == 0 else None
+    # consensus
+    out["consensus_ok"] = kf.consensus_bytes(b"same-bytes", "c1")
+    out["consensus_diff"] = kf.consensus_bytes(
+        ("val%d" % rank).encode(), "c2")
+    # bf16 sum
+    import torch
+
+    tb = torch.full((33,), 1.0 + rank, dtype=torch.bfloat16)
+    _core.all_reduce(tb.data_ptr(), tb.data_ptr(), 33, 9, 0, "bf")
+    out["bf16_sum"] = float(tb[0])
+    kf.finalize()
+    return out
+
+
+def p2p_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    blob = np_.full(1000, float(rank), dtype=np_.float32)
+    _core.save("model", blob.ctypes.data, blob.nbytes)
+    kf.barrier()
+    target = (rank + 1) % np
+    got = np_.zeros_like(blob)
+    ok = _core.request(target, "model", got.ctypes.data, got.nbytes)
+    assert ok, "request failed"
+    assert float(got[0]) == float(target)
+    kf.barrier()
+    kf.finalize()
+    return True
+
+
+def hierarchical_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    x = np_.full(130_000, 1.0, dtype=np_.float32)
+    _core.local_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "lr")
+    _core.cross_all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0,
+                           "cx")
+    _core.local_broadcast(x.ctypes.data, x.ctypes.data, x.size, 10, "lb")
+    kf.barrier()
+    kf.finalize()
+    return float(x[0])
+
+
+def settree_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    # install a chain 0 <- 1 <- 2 ... as the topology
+    parent = [max(0, i - 1) for i in range(np)]
+    _core.set_tree(parent)
+    x = np_.full(5, 1.0, dtype=np_.float32)
+    _core.all_reduce(x.ctypes.data, x.ctypes.data, 5, 10, 0, "t")
+    kf.finalize()
+    return float(x[0])
+
+
+def monitoring_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    x = np_.ones(100_000, dtype=np_.float32)
+    for _ in range(3):
+        _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "m")
+    stats = _core.strategy_stats()
+    lat = _core.peer_latencies_us()
+    eg = _core.egress_bytes()
+    interference = _core.check_interference(0.8)
+    kf.finalize()
+    total_ops = sum(s["ops"] for s in stats)
+    return {
+        "ops": total_ops,
+        "lat_len": len(lat),
+        "egress_nonzero": any(v > 0 for v in eg.values()),
+        "interference": interference,
+    }

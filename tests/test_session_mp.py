@@ -1,0 +1,55 @@
+"""Multi-process collective engine tests over loopback (the reference's
+core pattern: real runtime, N local processes, no GPUs —
+scripts/tests/run-integration-tests.sh sweeps np x strategies)."""
+import pytest
+
+from mp_helpers import (allreduce_body, hierarchical_body, monitoring_body,
+                        p2p_body, settree_body, spawn_cluster)
+
+
+@pytest.mark.parametrize("strategy",
+                         ["STAR", "RING", "BINARY_TREE", "CLIQUE",
+                          "BINARY_TREE_STAR"])
+@pytest.mark.parametrize("np", [2, 4])
+def test_allreduce_strategies(np, strategy, port_block):
+    results = spawn_cluster(allreduce_body, np, port_block, strategy)
+    expect_sum = sum(range(1, np + 1))
+    for r in results:
+        assert r["small_sum"] == pytest.approx(expect_sum)
+        assert r["min"] == 5
+        assert r["max"] == (np - 1) * 10 + 5
+        assert r["bcast"] == 0.0
+        assert r["gathered"] == list(range(np))
+        assert r["consensus_ok"] is True
+        assert r["consensus_diff"] is False
+        assert r["bf16_sum"] == pytest.approx(
+            sum(1.0 + i for i in range(np)))
+    assert results[0]["rooted"] == list(range(np))
+
+
+def test_allreduce_np1(port_block):
+    results = spawn_cluster(allreduce_body, 1, port_block, "AUTO")
+    assert results[0]["small_sum"] == 1.0
+
+
+def test_p2p_store(port_block):
+    assert spawn_cluster(p2p_body, 3, port_block) == [True, True, True]
+
+
+def test_hierarchical(port_block):
+    results = spawn_cluster(hierarchical_body, 4, port_block)
+    assert all(r == pytest.approx(4.0) for r in results)
+
+
+def test_set_tree(port_block):
+    results = spawn_cluster(settree_body, 4, port_block)
+    assert all(r == pytest.approx(4.0) for r in results)
+
+
+def test_monitoring(port_block):
+    results = spawn_cluster(monitoring_body, 2, port_block)
+    for r in results:
+        assert r["ops"] == 3
+        assert r["lat_len"] == 2
+        assert r["egress_nonzero"]
+        assert r["interference"] is False
