@@ -1,0 +1,200 @@
+"""Flagship benchmark: ResNet-50 Sync-SGD images/sec on MI355X.
+
+Matches BASELINE.json: synthetic ImageNet-shaped data, random-init
+ResNet-50, per-GPU batch 64, bf16, S-SGD gradient all-reduce over RCCL/xGMI
+(the reference's benchmark config: benchmarks/system/benchmark_kungfu.py
+with tf.keras ResNet50, batch 64/GPU).
+
+Launch:
+  python bench.py --gpus 1 --steps 30 --warmup 10
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+Rank 0 prints ONE JSON line with the whole-job images/sec (max step time
+over ranks).
+"""
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch-size", type=int, default=64)
+    p.add_argument("--model", default="resnet50",
+                   choices=["resnet50", "bert", "slp"])
+    p.add_argument("--optimizer", default="sync",
+                   choices=["sync", "sma", "pair", "gns"])
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--bucket-mb", type=int, default=32)
+    p.add_argument("--no-overlap", action="store_true")
+    p.add_argument("--channels-last", dest="channels_last",
+                   action="store_true", default=True)
+    return p.parse_args()
+
+
+def build_model_and_data(args, device, dtype):
+    if args.model == "resnet50":
+        from kungfu_amd.models import resnet50
+
+        model = resnet50()
+        x = torch.randn(args.batch_size, 3, 224, 224)
+        y = torch.randint(0, 1000, (args.batch_size,), device=device)
+        model = model.to(device=device, dtype=dtype)
+        x = x.to(device=device, dtype=dtype)
+        if args.channels_last and device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+            x = x.contiguous(memory_format=torch.channels_last)
+
+        def step_fn(opt):
+            opt.zero_grad()
+            out = model(x)
+            loss = torch.nn.functional.cross_entropy(out.float(), y)
+            loss.backward()
+            opt.step()
+            return loss
+
+        return model, step_fn, args.batch_size
+    if args.model == "bert":
+        from kungfu_amd.models import bert_base
+
+        model = bert_base(max_len=max(args.seq_len, 128)).to(
+            device=device, dtype=dtype)
+        ids = torch.randint(0, 30522, (args.batch_size, args.seq_len),
+                            device=device)
+        labels = torch.randint(0, 30522,
+                               (args.batch_size, args.seq_len),
+                               device=device)
+
+        def step_fn(opt):
+            opt.zero_grad()
+            out = model(ids)
+            loss = torch.nn.functional.cross_entropy(
+                out.float().flatten(0, 1), labels.flatten())
+            loss.backward()
+            opt.step()
+            return loss
+
+        return model, step_fn, args.batch_size
+    from kungfu_amd.models import SLP
+
+    model = SLP().to(device=device, dtype=dtype)
+    x = torch.randn(args.batch_size, 1, 28, 28, device=device, dtype=dtype)
+    y = torch.randint(0, 10, (args.batch_size,), device=device)
+
+    def step_fn(opt):
+        opt.zero_grad()
+        loss = torch.nn.functional.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+        return loss
+
+    return model, step_fn, args.batch_size
+
+
+def wrap_optimizer(args, model):
+    from kungfu_amd import optimizers as kfo
+
+    inner = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    if args.optimizer == "sync":
+        return kfo.SynchronousSGDOptimizer(
+            inner, bucket_bytes=args.bucket_mb << 20,
+            overlap=not args.no_overlap)
+    if args.optimizer == "sma":
+        return kfo.SynchronousAveragingOptimizer(inner)
+    if args.optimizer == "pair":
+        return kfo.PairAveragingOptimizer(inner)
+    return kfo.MonitorGradientNoiseScaleOptimizer(
+        inner, device_batch_size=args.batch_size)
+
+
+def max_over_ranks(value):
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    if kf.size() == 1:
+        return value
+    buf = np.array([value], dtype=np.float64)
+    out = np.zeros(1, dtype=np.float64)
+    _core.all_reduce(buf.ctypes.data, out.ctypes.data, 1, 11, 2, "|benchmax")
+    return float(out[0])
+
+
+def main():
+    args = parse_args()
+    import kungfu_amd as kf
+
+    kf.init()
+    rank, world = kf.rank(), kf.size()
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda:%d" % torch.cuda.current_device()
+                          if use_cuda else "cpu")
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+
+    torch.manual_seed(1234 + rank)
+    model, step_fn, per_gpu_batch = build_model_and_data(args, device,
+                                                         dtype)
+    from kungfu_amd.ops import broadcast_model
+
+    broadcast_model(model)
+    opt = wrap_optimizer(args, model)
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step_fn(opt)
+    sync()
+    kf.run_barrier()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step_fn(opt)
+    sync()
+    kf.run_barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
+    elapsed = max_over_ranks(elapsed)
+
+    n_gpus = world if use_cuda else world  # one rank per GPU
+    total_items = args.steps * per_gpu_batch * world
+    value = total_items / elapsed
+    unit = "images/sec" if args.model != "bert" else "sequences/sec"
+    if rank == 0:
+        result = {
+            "metric": "images/sec" if args.model == "resnet50" else
+                      ("%s %s" % (args.model, unit)),
+            "value": round(value, 2),
+            "unit": unit,
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": per_gpu_batch * world,
+                "seq_len": args.seq_len if args.model == "bert" else None,
+                "parallelism": "dp%d" % world,
+                "optimizer": args.optimizer,
+            },
+        }
+        print(json.dumps(result), flush=True)
+    kf.finalize()
+
+
+if __name__ == "__main__":
+    main()

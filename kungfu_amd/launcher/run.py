@@ -142,8 +142,9 @@ class Runner:
                  if s.rsplit(":", 1)[0] == ip]
         if spec in local:
             slot = local.index(spec)
-            env.setdefault("HIP_VISIBLE_DEVICES", str(slot))
-            env.setdefault("CUDA_VISIBLE_DEVICES", str(slot))
+            for key in ("HIP_VISIBLE_DEVICES", "CUDA_VISIBLE_DEVICES"):
+                if not env.get(key):  # empty or unset: assign our slot
+                    env[key] = str(slot)
         env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
         return env
 

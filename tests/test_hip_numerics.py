@@ -1,0 +1,122 @@
+"""Numerics of the gfx950 HIP kernels vs plain PyTorch fp32 references.
+
+Each kernel result is compared against the same op computed by torch in
+fp32 on the same data (tolerances account for bf16 rounding)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from kungfu_amd.ops import hip as hip_ops
+else:  # collected on CPU boxes but skipped
+    hip_ops = None
+
+
+def _rand(n, dtype):
+    return (torch.rand(n, device="cuda", dtype=torch.float32) - 0.5).to(
+        dtype)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16,
+                                   torch.float16])
+@pytest.mark.parametrize("n", [1, 63, 4096, 1 << 20])
+def test_avg_inplace(dtype, n):
+    y = _rand(n, dtype)
+    x = _rand(n, dtype)
+    ref = (0.7 * y.float() + 0.3 * x.float())
+    hip_ops.avg_inplace(y, x, alpha=0.3)
+    torch.cuda.synchronize()
+    tol = 1e-6 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(y.float(), ref, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("n", [17, 100_000, (1 << 22) + 5])
+def test_norm2(dtype, n):
+    x = _rand(n, dtype)
+    ref = float(x.float().pow(2).sum())
+    out = hip_ops.norm2(x)
+    torch.cuda.synchronize()
+    assert float(out.item()) == pytest.approx(ref, rel=1e-3)
+
+
+def test_dot():
+    x = _rand(1 << 20, torch.bfloat16)
+    y = _rand(1 << 20, torch.bfloat16)
+    ref = float((x.float() * y.float()).sum())
+    out = hip_ops.dot(x, y)
+    torch.cuda.synchronize()
+    assert float(out.item()) == pytest.approx(ref, rel=1e-2, abs=1.0)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fusion_pack_unpack(dtype):
+    torch.manual_seed(0)
+    sizes = [1, 7, 64, 1000, 4096, 1 << 18, 3]
+    tensors = [_rand(s, dtype) for s in sizes]
+    align = 64
+    offsets, off = [], 0
+    for t in tensors:
+        offsets.append(off)
+        off += (t.numel() + align - 1) // align * align
+    plan = hip_ops.FusionPlan(tensors, offsets, dtype)
+    fused = torch.zeros(off, device="cuda", dtype=dtype)
+    plan.pack(fused)
+    torch.cuda.synchronize()
+    for t, o in zip(tensors, offsets):
+        assert torch.equal(fused[o:o + t.numel()], t)
+    # unpack with scale into the segment tensors
+    fused2 = fused * 2
+    plan2 = hip_ops.FusionPlan(tensors, offsets, dtype)
+    plan2.unpack(fused2.contiguous(), scale=0.5)
+    torch.cuda.synchronize()
+    for t, o in zip(tensors, offsets):
+        ref = (fused2[o:o + t.numel()].float() * 0.5).to(dtype)
+        assert torch.allclose(t.float(), ref.float(), atol=1e-2, rtol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize("nesterov", [False, True])
+def test_sgd_momentum(dtype, nesterov):
+    n = 100_000
+    torch.manual_seed(1)
+    p = _rand(n, dtype)
+    g = _rand(n, dtype)
+    m = torch.rand(n, device="cuda", dtype=torch.float32)
+    # fp32 reference of the same update rule
+    lr, mu, wd, gs = 0.1, 0.9, 1e-4, 0.5
+    pf, gf, mf = p.float().clone(), g.float().clone(), m.clone()
+    gi = gf * gs + wd * pf
+    mf2 = mu * mf + gi
+    upd = gi + mu * mf2 if nesterov else mf2
+    ref_p = pf - lr * upd
+    hip_ops.sgd_momentum(p, g, m, lr=lr, momentum=mu, weight_decay=wd,
+                         grad_scale=gs, nesterov=nesterov)
+    torch.cuda.synchronize()
+    tol = 1e-6 if dtype == torch.float32 else 2e-2
+    assert torch.allclose(p.float(), ref_p, atol=tol, rtol=tol)
+    assert torch.allclose(m, mf2, atol=1e-5, rtol=1e-5)
+
+
+@pytest.mark.parametrize("op,fn", [
+    ("sum", lambda a, b: a + b),
+    ("min", torch.minimum),
+    ("max", torch.maximum),
+    ("prod", lambda a, b: a * b),
+])
+def test_transform2(op, fn):
+    z = _rand(65537, torch.float32)
+    x = _rand(65537, torch.float32)
+    ref = fn(z.clone(), x)
+    hip_ops.transform2(z, x, op=op)
+    torch.cuda.synchronize()
+    assert torch.allclose(z, ref, atol=1e-6)
+
+
+def test_scale():
+    y = _rand(12345, torch.bfloat16)
+    ref = (y.float() * 0.125).to(torch.bfloat16)
+    hip_ops.scale_(y, 0.125)
+    torch.cuda.synchronize()
+    assert torch.equal(y, ref)

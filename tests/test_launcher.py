@@ -1,0 +1,52 @@
+"""kungfu-run launcher end-to-end on loopback (reference pattern:
+run-integration-tests.sh / run-train-tests.sh under the real launcher)."""
+import os
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_launcher(args, timeout=180):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    return subprocess.run(
+        [sys.executable, "-m", "kungfu_amd.run"] + args,
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=timeout)
+
+
+def test_kungfu_run_np2_mnist_slp(port_block):
+    r = run_launcher([
+        "-np", "2", "-port", str(port_block), "-port-range",
+        str(port_block + 1), "-timeout", "150s",
+        sys.executable, "examples/mnist_slp.py", "--n-epochs", "2",
+    ])
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert r.stdout.count("FINAL") == 2
+
+
+def test_kungfu_run_env_protocol(port_block):
+    script = ("import kungfu_amd as kf, os; kf.init(with_torch=False); "
+              "print('R', kf.rank(), kf.size(), kf.local_rank(), "
+              "os.environ.get('HIP_VISIBLE_DEVICES'))")
+    r = run_launcher([
+        "-np", "3", "-port", str(port_block), "-port-range",
+        str(port_block + 1), "-strategy", "RING",
+        sys.executable, "-c", script,
+    ])
+    assert r.returncode == 0, r.stdout + r.stderr
+    import re
+
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", r.stdout)
+    lines = sorted(ln.split("] ", 1)[1].strip()
+                   for ln in plain.splitlines() if "] R " in ln)
+    assert lines == ["R 0 3 0 0", "R 1 3 1 1", "R 2 3 2 2"], repr(plain)
+
+
+def test_kungfu_run_propagates_failure(port_block):
+    r = run_launcher([
+        "-np", "2", "-port", str(port_block), "-port-range",
+        str(port_block + 1),
+        sys.executable, "-c", "import sys; sys.exit(3)",
+    ])
+    assert r.returncode == 3
