@@ -36,8 +36,8 @@ def parse_args():
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     p.add_argument("--bucket-mb", type=int, default=32)
     p.add_argument("--no-overlap", action="store_true")
-    p.add_argument("--channels-last", dest="channels_last",
-                   action="store_true", default=True)
+    p.add_argument("--channels-last", dest="channels_last", type=int,
+                   default=0)  # measured: NCHW beats NHWC on gfx950 MIOpen
     return p.parse_args()
 
 
@@ -130,6 +130,7 @@ def max_over_ranks(value):
 
 def main():
     args = parse_args()
+    torch.backends.cudnn.benchmark = True  # MIOpen find once per shape
     import kungfu_amd as kf
 
     kf.init()

@@ -1,0 +1,73 @@
+"""Localize ResNet-50 single-GPU step cost: fwd / fwd+bwd / full step,
+NCHW vs channels_last, find-mode sensitivity. Prints ms per phase."""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def timeit(fn, steps=10, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / steps * 1000
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--dtype", default="bf16")
+    p.add_argument("--channels-last", type=int, default=1)
+    p.add_argument("--benchmark-mode", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    args = p.parse_args()
+
+    torch.backends.cudnn.benchmark = bool(args.benchmark_mode)
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    from kungfu_amd.models import resnet50
+
+    model = resnet50().to("cuda", dtype)
+    x = torch.randn(args.batch, 3, 224, 224, device="cuda", dtype=dtype)
+    y = torch.randint(0, 1000, (args.batch,), device="cuda")
+    if args.channels_last:
+        model = model.to(memory_format=torch.channels_last)
+        x = x.contiguous(memory_format=torch.channels_last)
+
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9,
+                          foreach=True)
+
+    def fwd():
+        with torch.no_grad():
+            model(x)
+
+    def fwd_bwd():
+        for prm in model.parameters():
+            prm.grad = None
+        loss = torch.nn.functional.cross_entropy(model(x).float(), y)
+        loss.backward()
+
+    def full():
+        opt.zero_grad(set_to_none=False)
+        loss = torch.nn.functional.cross_entropy(model(x).float(), y)
+        loss.backward()
+        opt.step()
+
+    t_fwd = timeit(fwd, args.steps)
+    t_fb = timeit(fwd_bwd, args.steps)
+    t_full = timeit(full, args.steps)
+    print("batch=%d dtype=%s cl=%d bench=%d | fwd=%.1fms fwd+bwd=%.1fms "
+          "full=%.1fms opt=%.1fms img/s=%.0f" %
+          (args.batch, args.dtype, args.channels_last, args.benchmark_mode,
+           t_fwd, t_fb, t_full, t_full - t_fb, args.batch / t_full * 1000))
+
+
+if __name__ == "__main__":
+    main()

@@ -93,12 +93,19 @@ class RunnerServer {
     // returns (name, payload) or None after timeout
     py::object poll(int timeout_ms)
     {
-        std::unique_lock<std::mutex> lk(mu_);
-        if (!cv_.wait_for(lk, std::chrono::milliseconds(timeout_ms),
-                          [this] { return !q_.empty(); }))
-            return py::none();
-        auto item = q_.front();
-        q_.pop_front();
+        std::pair<std::string, std::string> item;
+        bool found = false;
+        {
+            py::gil_scoped_release rel;  // wait without the GIL
+            std::unique_lock<std::mutex> lk(mu_);
+            if (cv_.wait_for(lk, std::chrono::milliseconds(timeout_ms),
+                             [this] { return !q_.empty(); })) {
+                item = std::move(q_.front());
+                q_.pop_front();
+                found = true;
+            }
+        }
+        if (!found) return py::none();
         return py::make_tuple(item.first, py::bytes(item.second));
     }
     void stop() { server_->stop(); }
@@ -290,8 +297,7 @@ PYBIND11_MODULE(_core, m)
     py::class_<RunnerServer>(m, "RunnerServer")
         .def(py::init<const std::string &, bool>(), py::arg("self_spec"),
              py::arg("use_unix") = false)
-        .def("poll", &RunnerServer::poll, py::arg("timeout_ms"),
-             py::call_guard<py::gil_scoped_release>())
+        .def("poll", &RunnerServer::poll, py::arg("timeout_ms"))
         .def("stop", &RunnerServer::stop,
              py::call_guard<py::gil_scoped_release>());
 

@@ -30,6 +30,17 @@ DEFAULT_BUCKET_BYTES = 32 << 20
 _ALIGN = 64  # element alignment of bucket slices (16B vector loads)
 
 
+def _alias_view(flat_slice, p):
+    """View a flat-buffer slice with the parameter's memory layout, so
+    autograd's accumulate-grad writes straight into the bucket without a
+    layout repack (the 'gradient layout contract': conv weights held
+    channels_last want channels_last grads)."""
+    if p.dim() == 4 and p.is_contiguous(memory_format=torch.channels_last):
+        n, c, h, w = p.shape
+        return flat_slice.view(n, h, w, c).permute(0, 3, 1, 2)
+    return flat_slice.view_as(p)
+
+
 class _Bucket:
     __slots__ = ("index", "params", "flat", "numel", "ready", "work",
                  "launched")
@@ -92,7 +103,7 @@ class GradBucketReducer:
                 if p.dtype != dtype or p.device != dev:
                     raise ValueError(
                         "mixed dtype/device parameters in one reducer")
-                view = b.flat[off:off + p.numel()].view_as(p)
+                view = _alias_view(b.flat[off:off + p.numel()], p)
                 if p.grad is not None:
                     view.copy_(p.grad)  # preserve grads on mid-run adoption
                 p.grad = view
@@ -205,7 +216,7 @@ class FlatParamGroup:
             self._plan.pack(self.flat)
         else:
             for p, off in zip(self.params, self.offsets):
-                self.flat[off:off + p.numel()].copy_(p.data.view(-1))
+                self.flat[off:off + p.numel()].copy_(p.data.reshape(-1))
         return self.flat
 
     def unpack(self, scale=1.0):
@@ -213,7 +224,7 @@ class FlatParamGroup:
             self._plan.unpack(self.flat, scale)
         else:
             for p, off in zip(self.params, self.offsets):
-                src = self.flat[off:off + p.numel()].view_as(p.data)
+                src = self.flat[off:off + p.numel()].view(p.shape)
                 if scale != 1.0:
                     p.data.copy_(src * scale)
                 else:

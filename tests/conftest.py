@@ -5,7 +5,9 @@ import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
-_PORT_COUNTER = [36000]
+import random as _random
+
+_PORT_COUNTER = [_random.randrange(20000, 60000, 64)]
 
 
 def pytest_configure(config):

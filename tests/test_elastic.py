@@ -1,0 +1,68 @@
+"""Elastic resize end-to-end: kungfu-run watch mode + builtin config server
++ schedule-driven resize (reference: test_tensorflow_resize.py /
+test_step_based_schedule.py under `kungfu-run -w`)."""
+import os
+import re
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def run_watch(args, timeout=240):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    return subprocess.run(
+        [sys.executable, "-m", "kungfu_amd.run"] + args,
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=timeout)
+
+
+def _plain(s):
+    return re.sub(r"\x1b\[[0-9;]*m", "", s)
+
+
+def test_elastic_grow_2_to_3(port_block):
+    r = run_watch([
+        "-np", "2", "-w", "-port", str(port_block), "-port-range",
+        str(port_block + 2), "-builtin-config-port", str(port_block + 1),
+        sys.executable, "examples/elastic_trainer.py",
+        "--schedule", "3:3", "--max-step", "8",
+    ])
+    out = _plain(r.stdout)
+    assert r.returncode == 0, out + _plain(r.stderr)
+    assert out.count("JOIN") == 3  # 2 initial + 1 joiner
+    resized = [ln for ln in out.splitlines() if "RESIZED" in ln]
+    assert any("size=3" in ln for ln in resized)
+    done = [ln for ln in out.splitlines() if "DONE" in ln]
+    assert len(done) == 3
+    assert all("size=3 step=8" in ln for ln in done)
+
+
+def test_elastic_shrink_3_to_2(port_block):
+    r = run_watch([
+        "-np", "3", "-w", "-port", str(port_block), "-port-range",
+        str(port_block + 2), "-builtin-config-port", str(port_block + 1),
+        sys.executable, "examples/elastic_trainer.py",
+        "--schedule", "2:2", "--max-step", "6",
+    ])
+    out = _plain(r.stdout)
+    assert r.returncode == 0, out + _plain(r.stderr)
+    assert out.count("DETACHED") == 1
+    done = [ln for ln in out.splitlines() if "DONE" in ln]
+    assert len(done) == 2
+    assert all("size=2 step=6" in ln for ln in done)
+
+
+def test_elastic_grow_and_shrink(port_block):
+    r = run_watch([
+        "-np", "2", "-w", "-port", str(port_block), "-port-range",
+        str(port_block + 2), "-builtin-config-port", str(port_block + 1),
+        sys.executable, "examples/elastic_trainer.py",
+        "--schedule", "2:4,5:2", "--max-step", "8",
+    ], timeout=300)
+    out = _plain(r.stdout)
+    assert r.returncode == 0, out + _plain(r.stderr)
+    done = [ln for ln in out.splitlines() if "DONE" in ln]
+    assert len(done) == 2
+    assert all("size=2 step=8" in ln for ln in done)
+    assert out.count("DETACHED") == 2
