@@ -33,15 +33,27 @@ def parse_args():
     p.add_argument("--optimizer", default="sync",
                    choices=["sync", "sma", "pair", "gns"])
     p.add_argument("--seq-len", type=int, default=128)
-    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--dtype", default="bf16",
+                   choices=["bf16", "bf16-pure", "fp32"],
+                   help="bf16 = fp32 master weights + autocast bf16 compute"
+                        " (fastest measured); bf16-pure = all-bf16 model")
     p.add_argument("--bucket-mb", type=int, default=32)
     p.add_argument("--no-overlap", action="store_true")
     p.add_argument("--channels-last", dest="channels_last", type=int,
-                   default=0)  # measured: NCHW beats NHWC on gfx950 MIOpen
+                   default=None,
+                   help="default: on for autocast-bf16 (NHWC igemm convs, "
+                        "no transposes: 4914 vs 3581 img/s), off otherwise")
     return p.parse_args()
 
 
-def build_model_and_data(args, device, dtype):
+def build_model_and_data(args, device, dtype, amp):
+    import contextlib
+
+    def amp_ctx():
+        if amp and device.type == "cuda":
+            return torch.autocast("cuda", dtype=torch.bfloat16)
+        return contextlib.nullcontext()
+
     if args.model == "resnet50":
         from kungfu_amd.models import resnet50
 
@@ -56,7 +68,8 @@ def build_model_and_data(args, device, dtype):
 
         def step_fn(opt):
             opt.zero_grad()
-            out = model(x)
+            with amp_ctx():
+                out = model(x)
             loss = torch.nn.functional.cross_entropy(out.float(), y)
             loss.backward()
             opt.step()
@@ -68,6 +81,8 @@ def build_model_and_data(args, device, dtype):
 
         model = bert_base(max_len=max(args.seq_len, 128)).to(
             device=device, dtype=dtype)
+        if amp:
+            model = model.to(dtype=torch.float32)
         ids = torch.randint(0, 30522, (args.batch_size, args.seq_len),
                             device=device)
         labels = torch.randint(0, 30522,
@@ -76,7 +91,8 @@ def build_model_and_data(args, device, dtype):
 
         def step_fn(opt):
             opt.zero_grad()
-            out = model(ids)
+            with amp_ctx():
+                out = model(ids)
             loss = torch.nn.functional.cross_entropy(
                 out.float().flatten(0, 1), labels.flatten())
             loss.backward()
@@ -138,11 +154,14 @@ def main():
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda:%d" % torch.cuda.current_device()
                           if use_cuda else "cpu")
-    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
+    amp = args.dtype == "bf16"
+    dtype = torch.bfloat16 if args.dtype == "bf16-pure" else torch.float32
+    if args.channels_last is None:
+        args.channels_last = 1 if amp else 0
 
     torch.manual_seed(1234 + rank)
     model, step_fn, per_gpu_batch = build_model_and_data(args, device,
-                                                         dtype)
+                                                         dtype, amp)
     from kungfu_amd.ops import broadcast_model
 
     broadcast_model(model)
@@ -183,7 +202,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": args.dtype,
+            "dtype": "bf16" if args.dtype != "fp32" else "fp32",
             "data": "synthetic",
             "config": {
                 "model": args.model,
