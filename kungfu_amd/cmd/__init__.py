@@ -12,6 +12,10 @@ import urllib.request
 from kungfu_amd.launcher.monitor import MONITOR_PORT
 
 
+def _monitor_port():
+    return int(os.environ.get("KUNGFU_MONITOR_PORT", MONITOR_PORT))
+
+
 def _monitor_host():
     runners = os.environ.get("KUNGFU_INIT_RUNNERS", "")
     if runners:
@@ -31,7 +35,7 @@ def _rank():
 def _send(key):
     try:
         req = urllib.request.Request(
-            "http://%s:%d/" % (_monitor_host(), MONITOR_PORT),
+            "http://%s:%d/" % (_monitor_host(), _monitor_port()),
             data=json.dumps({"key": key}).encode(), method="POST")
         urllib.request.urlopen(req, timeout=2)
         return True

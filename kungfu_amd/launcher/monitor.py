@@ -100,7 +100,7 @@ def _broadcast_otherdown(runner, min_epoch):
             continue
         try:
             req = urllib.request.Request(
-                "http://%s:%d/" % (ip, MONITOR_PORT),
+                "http://%s:%d/" % (ip, runner.args.monitor_port),
                 data=json.dumps(
                     {"key": "otherdown:%d" % min_epoch}).encode(),
                 method="POST")
@@ -129,7 +129,7 @@ def monitored_run(runner, grace=GRACE_SECONDS):
     (reference monitored.go:18-75)."""
     grace = grace or GRACE_SECONDS
     state = MonitorState()
-    srv = start_monitor_server(state)
+    srv = start_monitor_server(state, runner.args.monitor_port)
     world = len(runner.peers.split(","))
     try:
         attempt = 0
@@ -141,11 +141,11 @@ def monitored_run(runner, grace=GRACE_SECONDS):
                 runner.spawn(spec, peers_csv, runner.version)
             # supervise
             failed = False
+            code = 0
             while True:
                 with runner.lock:
                     live = list(runner.procs.items())
                 all_exited = True
-                code = 0
                 for spec, proc in live:
                     rc = proc.popen.poll()
                     if rc is None:

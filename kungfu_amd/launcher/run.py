@@ -47,6 +47,8 @@ def parse_args(argv=None):
     p.add_argument("-builtin-config-port", type=int, default=0)
     p.add_argument("-auto-recover", default=None,
                    help="enable failure detection + restart, e.g. '10s'")
+    p.add_argument("-monitor-port", type=int, default=7756,
+                   help="heartbeat monitor server port (auto-recover)")
     p.add_argument("-timeout", default=None, help="job timeout, e.g. '120s'")
     p.add_argument("-q", dest="quiet", action="store_true")
     p.add_argument("-logdir", default=None)
@@ -127,6 +129,12 @@ class Runner:
 
     def worker_env(self, spec, peers_csv, version):
         env = dict(os.environ)
+        # make the kungfu_amd package importable in workers regardless of
+        # how the launcher itself was invoked
+        pkg_root = os.path.dirname(
+            os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+        env["PYTHONPATH"] = pkg_root + os.pathsep + env.get("PYTHONPATH",
+                                                            "")
         env["KUNGFU_SELF_SPEC"] = spec
         env["KUNGFU_INIT_PEERS"] = peers_csv
         env["KUNGFU_INIT_RUNNERS"] = ",".join(
@@ -136,6 +144,7 @@ class Runner:
         if self.config_server_url():
             env["KUNGFU_CONFIG_SERVER"] = self.config_server_url()
         env["KUNGFU_JOB_START_TIMESTAMP"] = str(int(time.time()))
+        env["KUNGFU_MONITOR_PORT"] = str(self.args.monitor_port)
         # GPU slot assignment: local rank among this host's workers
         ip = spec.rsplit(":", 1)[0]
         local = [s for s in peers_csv.split(",")
@@ -322,8 +331,7 @@ class Runner:
         timeout = parse_duration(self.args.timeout)
         if self.args.delay:
             time.sleep(parse_duration(self.args.delay))
-        need_config = (self.args.watch or self.args.builtin_config_port
-                       or self.args.auto_recover)
+        need_config = self.args.watch or self.args.builtin_config_port
         if need_config and not self.args.config_server:
             self.start_builtin_config_server()
         if self.args.auto_recover:

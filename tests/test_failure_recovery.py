@@ -1,0 +1,31 @@
+"""Failure detection + auto-recovery (the fork's marquee feature):
+kungfu-run -auto-recover restarts crashed training with adjusted epochs
+(reference: runner/monitored.go + monitorserver/monitor.go)."""
+import os
+import re
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_auto_recover_restarts_after_crash(port_block, tmp_path):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    ckpt = str(tmp_path / "ckpt.pt")
+    r = subprocess.run(
+        [sys.executable, "-m", "kungfu_amd.run",
+         "-np", "2", "-port", str(port_block), "-port-range",
+         str(port_block + 1), "-auto-recover", "3s",
+         "-monitor-port", str(port_block + 60),
+         sys.executable, "examples/failure_recovery_trainer.py",
+         "--n-epochs", "4", "--crash-at-epoch", "2", "--ckpt", ckpt],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300)
+    out = re.sub(r"\x1b\[[0-9;]*m", "", r.stdout)
+    assert "CRASHING rank 0 now" in out
+    assert "failure detected" in out
+    assert out.count("RESTARTED from epoch 2") == 2
+    ends = [ln for ln in out.splitlines() if "TRAIN END" in ln]
+    assert len(ends) == 2, out
+    assert all("total_epochs=4" in ln for ln in ends)
+    assert r.returncode == 0, out + r.stderr
