@@ -1,5 +1,8 @@
 #include "endpoints.hpp"
 
+#include <unistd.h>
+
+#include <cstdio>
 #include <cstring>
 #include <stdexcept>
 
@@ -106,13 +109,59 @@ void CollectiveEndpoint::shutdown()
 
 // ---------- BlobStore ----------
 
+BlobStore::~BlobStore()
+{
+    // best-effort cleanup of shm mirrors
+    std::lock_guard<std::mutex> lk(mu_);
+    for (auto &kv : versions_) {
+        for (uint64_t v = kv.second; v > 0 && v + 2 > kv.second; --v) {
+            ::unlink(shm_file(kv.first, v).c_str());
+        }
+    }
+}
+
+std::string BlobStore::shm_file(const std::string &name, uint64_t ver) const
+{
+    std::string safe = name;
+    for (auto &c : safe) {
+        if (c == '/' || c == '.') c = '_';
+    }
+    return "/dev/shm/kungfu-amd-" + std::to_string(owner_port_) + "-" +
+           safe + ".v" + std::to_string(ver);
+}
+
 void BlobStore::save(const std::string &name, const void *data, size_t len)
 {
     auto blob = std::make_shared<const std::vector<uint8_t>>(
         (const uint8_t *)data, (const uint8_t *)data + len);
+    uint64_t ver;
+    {
+        std::lock_guard<std::mutex> lk(mu_);
+        blobs_[name] = std::move(blob);
+        ver = ++versions_[name];
+    }
+    // shm mirror: write-then-rename so readers only ever see whole blobs
+    std::string path = shm_file(name, ver);
+    std::string tmp = path + ".tmp";
+    FILE *f = std::fopen(tmp.c_str(), "wb");
+    if (f) {
+        bool ok = std::fwrite(data, 1, len, f) == len;
+        std::fclose(f);
+        if (ok && std::rename(tmp.c_str(), path.c_str()) == 0) {
+            std::lock_guard<std::mutex> lk(mu_);
+            shm_paths_[name] = path;
+        } else {
+            ::unlink(tmp.c_str());
+        }
+    }
+    if (ver > 2) ::unlink(shm_file(name, ver - 2).c_str());
+}
+
+std::string BlobStore::shm_path(const std::string &name) const
+{
     std::lock_guard<std::mutex> lk(mu_);
-    blobs_[name] = std::move(blob);
-    versions_[name]++;
+    auto it = shm_paths_.find(name);
+    return it == shm_paths_.end() ? std::string() : it->second;
 }
 
 std::shared_ptr<const std::vector<uint8_t>> BlobStore::get(
@@ -135,7 +184,18 @@ uint64_t BlobStore::version(const std::string &name) const
 void P2PEndpoint::on_frame(const PeerID &src, Frame &f)
 {
     if (f.flags & msgflag::IsRequest) {
-        // reply with our stored blob over our own client conn to src
+        // Colocated peers get a shm reference (memcpy-speed pull on one
+        // MI355X node); remote peers get the payload inline.
+        const bool colocated = src.ipv4 == self_.ipv4;
+        if (colocated) {
+            std::string path = store_.shm_path(f.name);
+            if (!path.empty()) {
+                client_.send(src, ConnType::P2P, f.name,
+                             msgflag::IsResponse | msgflag::ShmRef,
+                             path.data(), path.size());
+                return;
+            }
+        }
         auto blob = store_.get(f.name);
         if (blob) {
             client_.send(src, ConnType::P2P, f.name, msgflag::IsResponse,
@@ -159,6 +219,7 @@ void P2PEndpoint::on_frame(const PeerID &src, Frame &f)
         }
         std::lock_guard<std::mutex> lk(w->mu);
         w->failed = (f.flags & msgflag::RequestFailed) != 0;
+        w->shm_ref = (f.flags & msgflag::ShmRef) != 0;
         w->data = std::move(f.data);
         w->done = true;
         w->cv.notify_all();
@@ -192,6 +253,16 @@ bool P2PEndpoint::request(const PeerID &target, const std::string &name,
         return false;
     }
     if (w->failed) return false;
+    if (w->shm_ref) {
+        std::string path((const char *)w->data.data(), w->data.size());
+        FILE *f = std::fopen(path.c_str(), "rb");
+        if (!f) return false;
+        size_t got = std::fread(dst, 1, len, f);
+        // exactly len bytes and nothing more (size must match)
+        bool ok = got == len && std::fgetc(f) == EOF;
+        std::fclose(f);
+        return ok;
+    }
     if (w->data.size() != len) return false;
     std::memcpy(dst, w->data.data(), len);
     return true;

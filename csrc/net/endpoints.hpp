@@ -47,22 +47,35 @@ class CollectiveEndpoint {
     bool dead_ = false;
 };
 
-// In-memory blob store for P2P model exchange (reference: srcs/go/store/).
-// save() replaces the blob atomically via shared_ptr swap, so concurrent
-// request handlers keep a consistent snapshot without a window GC.
+// Blob store for P2P model exchange (reference: srcs/go/store/ versioned
+// store with a sliding version window, versionedstore.go:36-60).
+// save() replaces the in-memory blob atomically via shared_ptr swap AND
+// mirrors it into a versioned /dev/shm file, so colocated peers (the 8
+// workers of one MI355X node) pull models at memcpy speed instead of
+// through the loopback socket; the two most recent versions are kept on
+// disk so an in-flight reader never races a concurrent save.
 class BlobStore {
   public:
+    explicit BlobStore(uint16_t owner_port = 0) : owner_port_(owner_port) {}
+    ~BlobStore();
+    void set_owner_port(uint16_t port) { owner_port_ = port; }
     void save(const std::string &name, const void *data, size_t len);
     std::shared_ptr<const std::vector<uint8_t>> get(
         const std::string &name) const;
     uint64_t version(const std::string &name) const;
+    // Path of the current shm mirror ("" when shm is unavailable).
+    std::string shm_path(const std::string &name) const;
 
   private:
+    std::string shm_file(const std::string &name, uint64_t ver) const;
+
+    uint16_t owner_port_;
     mutable std::mutex mu_;
     std::unordered_map<std::string,
                        std::shared_ptr<const std::vector<uint8_t>>>
         blobs_;
     std::unordered_map<std::string, uint64_t> versions_;
+    std::unordered_map<std::string, std::string> shm_paths_;
 };
 
 // Request/response model pulls (AD-PSGD PairAveraging).
@@ -83,7 +96,7 @@ class P2PEndpoint {
         std::mutex mu;
         std::condition_variable cv;
         std::vector<uint8_t> data;
-        bool done = false, failed = false;
+        bool done = false, failed = false, shm_ref = false;
     };
     BlobStore &store_;
     Client &client_;

@@ -33,6 +33,7 @@ namespace msgflag {
 constexpr uint32_t IsResponse = 1u << 0;
 constexpr uint32_t RequestFailed = 1u << 1;
 constexpr uint32_t IsRequest = 1u << 2;
+constexpr uint32_t ShmRef = 1u << 3;  // payload is a /dev/shm path, not data
 }  // namespace msgflag
 
 struct Frame {

@@ -153,8 +153,17 @@ void Peer::start()
     started_ = true;
     workers_ = cfg_.init_peers;
     runners_ = cfg_.init_runners;
+    // aux subsystems driven by env (reference config/config.go:25-70)
+    if (!getenv_str("KUNGFU_USE_AFFINITY").empty()) {
+        bind_cpu_affinity(workers_.local_rank_of(cfg_.self),
+                          workers_.local_size_of(cfg_.self));
+    }
+    if (!getenv_str("KUNGFU_CONFIG_ENABLE_STALL_DETECTION").empty()) {
+        stall_ = std::make_unique<StallDetector>(3.0);
+    }
     client_ = std::make_unique<Client>(cfg_.self);
     client_->set_token(version_);
+    store_.set_owner_port(cfg_.self.port);
     if (!cfg_.single) {
         server_ = std::make_unique<Server>(cfg_.self, cfg_.use_unix);
         p2p_ = std::make_unique<P2PEndpoint>(store_, *client_, cfg_.self);
@@ -186,6 +195,22 @@ void Peer::start()
     if (r < 0) throw std::runtime_error("self not in init peer list");
     session_ = std::make_unique<Session>(workers_, r, *client_, collective_,
                                          cfg_.strategy);
+    if (!cfg_.single &&
+        !getenv_str("KUNGFU_CONFIG_ENABLE_MONITORING").empty()) {
+        // Prometheus-style counters at peer port + 10000
+        metrics_ = std::make_unique<MetricsServer>(
+            (uint16_t)(cfg_.self.port + 10000), [this] {
+                std::string out;
+                for (auto &kv : client_->egress_all()) {
+                    PeerID p;
+                    p.ipv4 = (uint32_t)(kv.first >> 16);
+                    p.port = (uint16_t)(kv.first & 0xffff);
+                    out += "kungfu_egress_bytes_total{peer=\"" + p.str() +
+                           "\"} " + std::to_string(kv.second) + "\n";
+                }
+                return out;
+            });
+    }
     // wait for all peers to be reachable, then an initial barrier
     if (!cfg_.single && workers_.size() > 1) {
         for (const auto &p : workers_.peers) {
@@ -202,6 +227,8 @@ void Peer::close()
 {
     if (!started_) return;
     started_ = false;
+    metrics_.reset();
+    stall_.reset();
     collective_.shutdown();
     if (p2p_) p2p_->shutdown();
     if (server_) server_->stop();

@@ -14,6 +14,7 @@
 #include "../net/endpoints.hpp"
 #include "../net/transport.hpp"
 #include "../session/session.hpp"
+#include "aux.hpp"
 
 namespace kf {
 
@@ -79,6 +80,7 @@ class Peer {
 
     // Monitoring
     std::vector<int64_t> peer_latencies_us();
+    StallDetector *stall_detector() { return stall_.get(); }
     std::map<uint64_t, uint64_t> egress_bytes() const
     {
         return client_->egress_all();
@@ -101,6 +103,8 @@ class Peer {
     BlobStore store_;
     std::unique_ptr<P2PEndpoint> p2p_;
     std::unique_ptr<Session> session_;
+    std::unique_ptr<StallDetector> stall_;
+    std::unique_ptr<MetricsServer> metrics_;
 };
 
 }  // namespace kf

@@ -31,6 +31,23 @@ Peer &peer()
     return *g_peer;
 }
 
+// RAII stall-detection guard around blocking collective calls
+struct StallGuard {
+    StallDetector *d = nullptr;
+    uint64_t id = 0;
+    explicit StallGuard(const char *what)
+    {
+        if (g_peer && g_peer->stall_detector()) {
+            d = g_peer->stall_detector();
+            id = d->enter(what);
+        }
+    }
+    ~StallGuard()
+    {
+        if (d) d->leave(id);
+    }
+};
+
 Workspace make_ws(uintptr_t send, uintptr_t recv, size_t count, int dtype,
                   int op, const std::string &name)
 {
@@ -154,6 +171,7 @@ PYBIND11_MODULE(_core, m)
     // ---- collectives (blocking; GIL released) ----
     m.def("barrier", [] {
         py::gil_scoped_release rel;
+        StallGuard sg("barrier");
         peer().session().barrier();
     });
     m.def("all_reduce",
@@ -161,6 +179,7 @@ PYBIND11_MODULE(_core, m)
              const std::string &name) {
               auto w = make_ws(s, r, count, dt, op, name);
               py::gil_scoped_release rel;
+              StallGuard sg("all_reduce");
               peer().session().all_reduce(w);
           });
     m.def("reduce",

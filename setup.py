@@ -28,6 +28,7 @@ CORE_SOURCES = [
     "csrc/net/endpoints.cpp",
     "csrc/session/session.cpp",
     "csrc/peer/peer.cpp",
+    "csrc/peer/aux.cpp",
     "csrc/pybind/module.cpp",
 ]
 
@@ -96,7 +97,7 @@ setup(
     packages=[
         "kungfu_amd", "kungfu_amd.ops", "kungfu_amd.optimizers",
         "kungfu_amd.parallel", "kungfu_amd.models", "kungfu_amd.utils",
-        "kungfu_amd.launcher", "kungfu_amd.cmd",
+        "kungfu_amd.launcher", "kungfu_amd.cmd", "kungfu_amd.benchmarks",
     ],
     ext_modules=[core_ext],
     cmdclass={"build_ext": BuildExt},

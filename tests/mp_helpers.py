@@ -194,3 +194,83 @@ def monitoring_body(rank, np):
         "egress_nonzero": any(v > 0 for v in eg.values()),
         "interference": interference,
     }
+
+
+def adaptive_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+    from kungfu_amd.parallel.adaptive import (check_interference_and_switch,
+                                              strategy_throughputs)
+
+    kf.init(with_torch=False)
+    x = np_.ones(200_000, dtype=np_.float32)
+    for _ in range(3):
+        _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "a")
+    stats = strategy_throughputs()
+    assert sum(s["ops"] for s in stats) == 3
+    # ratio=100 forces every peer to vote "interference" -> majority switch
+    before = _core.get_strategy()
+    for _ in range(2):  # first call records best, second must trip
+        new = check_interference_and_switch(ratio=100.0)
+        _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "a")
+        if new is not None:
+            break
+    after = _core.get_strategy()
+    # collectives still work after the switch on the new topology
+    y = np_.full(10, 1.0, dtype=np_.float32)
+    _core.all_reduce(y.ctypes.data, y.ctypes.data, 10, 10, 0, "post")
+    kf.finalize()
+    return {"before": before, "after": after, "sum": float(y[0])}
+
+
+def metrics_body(rank, np):
+    import urllib.request
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    x = np_.ones(100_000, dtype=np_.float32)
+    _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "m")
+    kf.barrier()
+    import os
+
+    port = int(os.environ["KUNGFU_SELF_SPEC"].rsplit(":", 1)[1]) + 10000
+    body = urllib.request.urlopen(
+        "http://127.0.0.1:%d/metrics" % port, timeout=5).read().decode()
+    kf.barrier()
+    kf.finalize()
+    return "kungfu_egress_bytes_total" in body
+
+
+def sampler_body(rank, np):
+    import kungfu_amd as kf
+    from kungfu_amd.data import ElasticShardSampler
+
+    kf.init(with_torch=False)
+    s = ElasticShardSampler(100, seed=1)
+    idx = list(s)
+    kf.finalize()
+    return idx
+
+
+def p2p_bigpull_body(rank, np):
+    import time
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    blob = np_.full(25_000_000, float(rank), dtype=np_.float32)  # 100 MB
+    _core.save("model", blob.ctypes.data, blob.nbytes)
+    kf.barrier()
+    got = np_.zeros_like(blob)
+    t0 = time.perf_counter()
+    ok = _core.request((rank + 1) % np, "model", got.ctypes.data,
+                       got.nbytes)
+    dt = time.perf_counter() - t0
+    assert ok and float(got[0]) == float((rank + 1) % np)
+    kf.barrier()
+    kf.finalize()
+    return blob.nbytes / dt / 1e9  # GB/s
