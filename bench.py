@@ -41,6 +41,9 @@ def parse_args():
     p.add_argument("--no-overlap", action="store_true")
     p.add_argument("--graph", type=int, default=1,
                    help="capture the step into a hipGraph and replay")
+    p.add_argument("--fused-bn", type=int, default=1,
+                   help="resnet50: use the gfx950 fused BN(+res+ReLU) "
+                        "kernels instead of autocast fp32 BN")
     p.add_argument("--channels-last", dest="channels_last", type=int,
                    default=None,
                    help="default: on for autocast-bf16 (NHWC igemm convs, "
@@ -59,7 +62,8 @@ def build_model_and_data(args, device, dtype, amp):
     if args.model == "resnet50":
         from kungfu_amd.models import resnet50
 
-        model = resnet50()
+        use_fused = bool(args.fused_bn) and amp and device.type == "cuda"
+        model = resnet50(fused_bn=use_fused)
         x = torch.randn(args.batch_size, 3, 224, 224)
         y = torch.randint(0, 1000, (args.batch_size,), device=device)
         model = model.to(device=device, dtype=dtype)

@@ -1,0 +1,386 @@
+// Fused BatchNorm(+residual)(+ReLU) kernels for NHWC bf16 activations,
+// fp32 parameters — gfx950.
+//
+// Why: under torch autocast the ResNet hot path runs BN in fp32 (cast
+// bf16->f32 copies + MIOpen spatial BN + separate residual-add and ReLU
+// elementwise kernels + f32->bf16 casts) — measured 5.2 ms of a 13 ms
+// ResNet-50 b64 step (profiles/resnet50_b64_bf16_steady_state_r01.md).
+// These kernels read/write bf16 exactly once per pass, accumulate in f32,
+// and fold residual-add + ReLU into the normalize pass.
+//
+// Layout: channels_last (NHWC): flat element i has channel i % C. All
+// kernels vectorize 8 consecutive channels per lane (16 B loads, G13) and
+// require C % 8 == 0 and C <= 8 * 256 (ResNet: 64..2048); the Python layer
+// falls back to eager BN otherwise.
+//
+// Per-channel reductions: each lane owns one channel-octet in registers
+// (f32x8 accumulators), reduces into LDS with shared-memory atomics, one
+// global atomic per channel per block (Guideline 12). With M = N*H*W in
+// the tens of thousands and <= 2048 blocks, global atomics are cold.
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace {
+
+constexpr int BLOCK = 256;
+
+typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+typedef __attribute__((ext_vector_type(4))) float float4v;
+
+__device__ inline float b2f(unsigned short u)
+{
+    union {
+        unsigned int i;
+        float f;
+    } v;
+    v.i = (unsigned int)u << 16;
+    return v.f;
+}
+
+__device__ inline unsigned short f2b(float f)
+{
+    union {
+        unsigned int i;
+        float f;
+    } v;
+    v.f = f;
+    unsigned int r = v.i + 0x7fff + ((v.i >> 16) & 1);  // round-nearest-even
+    return (unsigned short)(r >> 16);
+}
+
+inline int grid_for(long long work_items)
+{
+    long long g = (work_items + BLOCK - 1) / BLOCK;
+    if (g < 1) g = 1;
+    if (g > 2048) g = 2048;
+    return (int)g;
+}
+
+// ---- pass 1: per-channel sum / sumsq ----
+// x: [M][C] bf16; out sums: f32[2*C] {sum, sumsq} (pre-zeroed)
+__global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
+                                long long M, int C,
+                                float *__restrict__ sums)
+{
+    extern __shared__ float lds[];  // 2*C floats
+    const int gpr = C / 8;          // channel-octet groups per row
+    const int rows_per_blk = BLOCK / gpr;  // >= 1 given C <= 2048
+    const int g = threadIdx.x % gpr;
+    const int row_off = threadIdx.x / gpr;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) lds[i] = 0.f;
+    __syncthreads();
+
+    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (row_off < rows_per_blk) {
+        const long long row_step = (long long)gridDim.x * rows_per_blk;
+        for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+             r < M; r += row_step) {
+            const ushort8 v = *(const ushort8 *)(x + r * C + g * 8);
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                const float f = b2f(v[k]);
+                s[k] += f;
+                q[k] += f * f;
+            }
+        }
+    }
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        atomicAdd(&lds[g * 8 + k], s[k]);
+        atomicAdd(&lds[C + g * 8 + k], q[k]);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+        atomicAdd(&sums[i], lds[i]);
+    }
+}
+
+// ---- finalize: mean/var -> folded scale/shift + running stats ----
+// sums: {sum,sumsq}; outputs a = w*rstd, b = bias - mean*a; saves
+// mean/rstd for backward; updates running stats in place (momentum).
+__global__ void bn_finalize_kernel(const float *__restrict__ sums,
+                                   const float *__restrict__ weight,
+                                   const float *__restrict__ bias,
+                                   float *__restrict__ running_mean,
+                                   float *__restrict__ running_var,
+                                   float *__restrict__ save_mean,
+                                   float *__restrict__ save_rstd,
+                                   float *__restrict__ a,
+                                   float *__restrict__ b, long long M,
+                                   int C, float eps, float momentum)
+{
+    const int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    const float mean = sums[c] / (float)M;
+    const float var = fmaxf(sums[C + c] / (float)M - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    save_mean[c] = mean;
+    save_rstd[c] = rstd;
+    const float ac = weight[c] * rstd;
+    a[c] = ac;
+    b[c] = bias[c] - mean * ac;
+    if (momentum > 0.f) {
+        running_mean[c] += momentum * (mean - running_mean[c]);
+        const float unbiased =
+            M > 1 ? var * (float)M / (float)(M - 1) : var;
+        running_var[c] += momentum * (unbiased - running_var[c]);
+    }
+}
+
+// ---- pass 2: y = [relu]( a*x + b [+ res] ) ----
+template <bool RELU, bool RES>
+__global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
+                              const unsigned short *__restrict__ res,
+                              unsigned short *__restrict__ y,
+                              const float *__restrict__ a,
+                              const float *__restrict__ b, long long M,
+                              int C)
+{
+    const int gpr = C / 8;
+    const long long total = M * gpr;  // vec8 groups
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         j < total; j += stride) {
+        const int g = (int)(j % gpr);
+        const long long base = (j / gpr) * C + (long long)g * 8;
+        const ushort8 v = *(const ushort8 *)(x + base);
+        ushort8 rv;
+        if (RES) rv = *(const ushort8 *)(res + base);
+        ushort8 out;
+        const float4v a0 = *(const float4v *)(a + g * 8);
+        const float4v a1 = *(const float4v *)(a + g * 8 + 4);
+        const float4v b0 = *(const float4v *)(b + g * 8);
+        const float4v b1 = *(const float4v *)(b + g * 8 + 4);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            const float ak = k < 4 ? a0[k] : a1[k - 4];
+            const float bk = k < 4 ? b0[k] : b1[k - 4];
+            float u = fmaf(ak, b2f(v[k]), bk);
+            if (RES) u += b2f(rv[k]);
+            if (RELU) u = fmaxf(u, 0.f);
+            out[k] = f2b(u);
+        }
+        *(ushort8 *)(y + base) = out;
+    }
+}
+
+// ---- backward pass 1: per-channel sums of dy_m and dy_m * xhat ----
+// dy_m = dy * relu_mask (mask recomputed from x [+res]); xhat from
+// save_mean/save_rstd. out: f32[2*C] {sum_dy, sum_dyxhat} (pre-zeroed:
+// these ARE db and dw).
+template <bool RELU, bool RES>
+__global__ void bn_bwd_reduce_kernel(
+    const unsigned short *__restrict__ dy,
+    const unsigned short *__restrict__ x,
+    const unsigned short *__restrict__ res,
+    const float *__restrict__ a, const float *__restrict__ b,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    long long M, int C, float *__restrict__ sums)
+{
+    extern __shared__ float lds[];  // 2*C floats
+    const int gpr = C / 8;
+    const int rows_per_blk = BLOCK / gpr;
+    const int g = threadIdx.x % gpr;
+    const int row_off = threadIdx.x / gpr;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) lds[i] = 0.f;
+    __syncthreads();
+
+    float s1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float s2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    if (row_off < rows_per_blk) {
+        const long long row_step = (long long)gridDim.x * rows_per_blk;
+        for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+             r < M; r += row_step) {
+            const long long base = r * C + (long long)g * 8;
+            const ushort8 dv = *(const ushort8 *)(dy + base);
+            const ushort8 xv = *(const ushort8 *)(x + base);
+            ushort8 rv;
+            if (RES) rv = *(const ushort8 *)(res + base);
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                const int c = g * 8 + k;
+                const float xf = b2f(xv[k]);
+                float d = b2f(dv[k]);
+                if (RELU) {
+                    float u = fmaf(a[c], xf, b[c]);
+                    if (RES) u += b2f(rv[k]);
+                    d = u > 0.f ? d : 0.f;
+                }
+                const float xh = (xf - mean[c]) * rstd[c];
+                s1[k] += d;
+                s2[k] += d * xh;
+            }
+        }
+    }
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        atomicAdd(&lds[g * 8 + k], s1[k]);
+        atomicAdd(&lds[C + g * 8 + k], s2[k]);
+    }
+    __syncthreads();
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+        atomicAdd(&sums[i], lds[i]);
+    }
+}
+
+// ---- backward pass 2: dx (and d_res when fused residual) ----
+// dx = w*rstd * (dy_m - sum_dy/M - xhat * sum_dyxhat/M); d_res = dy_m.
+template <bool RELU, bool RES>
+__global__ void bn_bwd_dx_kernel(
+    const unsigned short *__restrict__ dy,
+    const unsigned short *__restrict__ x,
+    const unsigned short *__restrict__ res,
+    const float *__restrict__ a, const float *__restrict__ b,
+    const float *__restrict__ mean, const float *__restrict__ rstd,
+    const float *__restrict__ sums, long long M, int C,
+    unsigned short *__restrict__ dx, unsigned short *__restrict__ dres)
+{
+    const int gpr = C / 8;
+    const long long total = M * gpr;
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    const float invM = 1.f / (float)M;
+    for (long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         j < total; j += stride) {
+        const int g = (int)(j % gpr);
+        const long long base = (j / gpr) * C + (long long)g * 8;
+        const ushort8 dv = *(const ushort8 *)(dy + base);
+        const ushort8 xv = *(const ushort8 *)(x + base);
+        ushort8 rv;
+        if (RES) rv = *(const ushort8 *)(res + base);
+        ushort8 dxo, dro;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            const int c = g * 8 + k;
+            const float xf = b2f(xv[k]);
+            float d = b2f(dv[k]);
+            if (RELU) {
+                float u = fmaf(a[c], xf, b[c]);
+                if (RES) u += b2f(rv[k]);
+                d = u > 0.f ? d : 0.f;
+            }
+            if (RES) dro[k] = f2b(d);
+            const float xh = (xf - mean[c]) * rstd[c];
+            const float t =
+                d - sums[c] * invM - xh * (sums[C + c] * invM);
+            dxo[k] = f2b(a[c] * t);
+        }
+        *(ushort8 *)(dx + base) = dxo;
+        if (RES) *(ushort8 *)(dres + base) = dro;
+    }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// extern "C" launchers
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
+                       void *stream)
+{
+    if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
+    const int gpr = C / 8;
+    const int rows_per_blk = BLOCK / gpr;
+    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(bn_stats_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
+                       2 * C * sizeof(float), (hipStream_t)stream,
+                       (const unsigned short *)x, M, C, (float *)sums);
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_finalize(const void *sums, const void *weight,
+                          const void *bias, void *running_mean,
+                          void *running_var, void *save_mean,
+                          void *save_rstd, void *a, void *b, long long M,
+                          int C, float eps, float momentum, void *stream)
+{
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (const float *)sums,
+                       (const float *)weight, (const float *)bias,
+                       (float *)running_mean, (float *)running_var,
+                       (float *)save_mean, (float *)save_rstd, (float *)a,
+                       (float *)b, M, C, eps, momentum);
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
+                     const void *b, long long M, int C, int relu,
+                     void *stream)
+{
+    const long long total = M * (C / 8);
+    const dim3 grid(grid_for(total)), block(BLOCK);
+    const auto s = (hipStream_t)stream;
+#define CASE(R, E)                                                          \
+    hipLaunchKernelGGL((bn_fwd_kernel<R, E>), grid, block, 0, s,            \
+                       (const unsigned short *)x,                           \
+                       (const unsigned short *)res, (unsigned short *)y,    \
+                       (const float *)a, (const float *)b, M, C)
+    if (relu && res) CASE(true, true);
+    else if (relu) CASE(true, false);
+    else if (res) CASE(false, true);
+    else CASE(false, false);
+#undef CASE
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_bwd_reduce(const void *dy, const void *x, const void *res,
+                            const void *a, const void *b, const void *mean,
+                            const void *rstd, long long M, int C, int relu,
+                            void *sums, void *stream)
+{
+    if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
+    const int gpr = C / 8;
+    const int rows_per_blk = BLOCK / gpr;
+    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    const dim3 grid((uint32_t)blocks), block(BLOCK);
+    const auto s = (hipStream_t)stream;
+    const size_t lds = 2 * C * sizeof(float);
+#define CASE(R, E)                                                          \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, E>), grid, block, lds, s,   \
+                       (const unsigned short *)dy,                          \
+                       (const unsigned short *)x,                           \
+                       (const unsigned short *)res, (const float *)a,       \
+                       (const float *)b, (const float *)mean,               \
+                       (const float *)rstd, M, C, (float *)sums)
+    if (relu && res) CASE(true, true);
+    else if (relu) CASE(true, false);
+    else if (res) CASE(false, true);
+    else CASE(false, false);
+#undef CASE
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *res,
+                        const void *a, const void *b, const void *mean,
+                        const void *rstd, const void *sums, long long M,
+                        int C, int relu, void *dx, void *dres, void *stream)
+{
+    const long long total = M * (C / 8);
+    const dim3 grid(grid_for(total)), block(BLOCK);
+    const auto s = (hipStream_t)stream;
+#define CASE(R, E)                                                          \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<R, E>), grid, block, 0, s,         \
+                       (const unsigned short *)dy,                          \
+                       (const unsigned short *)x,                           \
+                       (const unsigned short *)res, (const float *)a,       \
+                       (const float *)b, (const float *)mean,               \
+                       (const float *)rstd, (const float *)sums, M, C,      \
+                       (unsigned short *)dx, (unsigned short *)dres)
+    if (relu && res) CASE(true, true);
+    else if (relu) CASE(true, false);
+    else if (res) CASE(false, true);
+    else CASE(false, false);
+#undef CASE
+    return hipGetLastError();
+}
+
+}  // extern "C"

@@ -17,6 +17,20 @@ namespace py = pybind11;
 
 extern "C" {
 hipError_t kf_pack(const void *, int, void *, int, void *);
+hipError_t kf_bn_stats(const void *, long long, int, void *, void *);
+hipError_t kf_bn_finalize(const void *, const void *, const void *, void *,
+                          void *, void *, void *, void *, void *, long long,
+                          int, float, float, void *);
+hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
+                     const void *, long long, int, int, void *);
+hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
+                            const void *, const void *, const void *,
+                            const void *, long long, int, int, void *,
+                            void *);
+hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
+                        const void *, const void *, const void *,
+                        const void *, const void *, long long, int, int,
+                        void *, void *, void *);
 hipError_t kf_unpack(const void *, int, const void *, float, int, void *);
 hipError_t kf_avg_inplace(void *, const void *, float, long long, int,
                           void *);
@@ -177,4 +191,58 @@ PYBIND11_MODULE(_hip, m)
                     "kf_transform2");
           });
     m.def("device_synchronize", [] { check(hipDeviceSynchronize(), "sync"); });
+
+    // ---- fused BatchNorm(+residual+ReLU), NHWC bf16 ----
+    m.def("bn_stats",
+          [](uintptr_t x, long long M, int C, uintptr_t sums,
+             uintptr_t stream) {
+              check(kf_bn_stats((const void *)x, M, C, (void *)sums,
+                                (void *)stream),
+                    "kf_bn_stats");
+          });
+    m.def("bn_finalize",
+          [](uintptr_t sums, uintptr_t w, uintptr_t bias, uintptr_t rmean,
+             uintptr_t rvar, uintptr_t smean, uintptr_t srstd, uintptr_t a,
+             uintptr_t b, long long M, int C, float eps, float momentum,
+             uintptr_t stream) {
+              check(kf_bn_finalize(
+                        (const void *)sums, (const void *)w,
+                        (const void *)bias, (void *)rmean, (void *)rvar,
+                        (void *)smean, (void *)srstd, (void *)a, (void *)b,
+                        M, C, eps, momentum, (void *)stream),
+                    "kf_bn_finalize");
+          });
+    m.def("bn_fwd",
+          [](uintptr_t x, uintptr_t res, uintptr_t y, uintptr_t a,
+             uintptr_t b, long long M, int C, bool relu, uintptr_t stream) {
+              check(kf_bn_fwd((const void *)x, (const void *)res,
+                              (void *)y, (const void *)a, (const void *)b,
+                              M, C, relu ? 1 : 0, (void *)stream),
+                    "kf_bn_fwd");
+          });
+    m.def("bn_bwd_reduce",
+          [](uintptr_t dy, uintptr_t x, uintptr_t res, uintptr_t a,
+             uintptr_t b, uintptr_t mean, uintptr_t rstd, long long M,
+             int C, bool relu, uintptr_t sums, uintptr_t stream) {
+              check(kf_bn_bwd_reduce((const void *)dy, (const void *)x,
+                                     (const void *)res, (const void *)a,
+                                     (const void *)b, (const void *)mean,
+                                     (const void *)rstd, M, C,
+                                     relu ? 1 : 0, (void *)sums,
+                                     (void *)stream),
+                    "kf_bn_bwd_reduce");
+          });
+    m.def("bn_bwd_dx",
+          [](uintptr_t dy, uintptr_t x, uintptr_t res, uintptr_t a,
+             uintptr_t b, uintptr_t mean, uintptr_t rstd, uintptr_t sums,
+             long long M, int C, bool relu, uintptr_t dx, uintptr_t dres,
+             uintptr_t stream) {
+              check(kf_bn_bwd_dx((const void *)dy, (const void *)x,
+                                 (const void *)res, (const void *)a,
+                                 (const void *)b, (const void *)mean,
+                                 (const void *)rstd, (const void *)sums, M,
+                                 C, relu ? 1 : 0, (void *)dx, (void *)dres,
+                                 (void *)stream),
+                    "kf_bn_bwd_dx");
+          });
 }

@@ -34,6 +34,7 @@ CORE_SOURCES = [
 
 HIP_SOURCES = [
     "csrc/hip/kernels.hip",
+    "csrc/hip/fused_bn.hip",
     "csrc/hip/module_hip.cpp",
 ]
 
