@@ -39,8 +39,9 @@ def parse_args():
                         " (fastest measured); bf16-pure = all-bf16 model")
     p.add_argument("--bucket-mb", type=int, default=32)
     p.add_argument("--no-overlap", action="store_true")
-    p.add_argument("--graph", type=int, default=1,
-                   help="capture the step into a hipGraph and replay")
+    p.add_argument("--graph", type=int, default=None,
+                   help="capture the step into a hipGraph and replay "
+                        "(default: on for single-GPU runs)")
     p.add_argument("--fused-bn", type=int, default=1,
                    help="resnet50: use the gfx950 fused BN(+res+ReLU) "
                         "kernels instead of autocast fp32 BN")
@@ -180,6 +181,8 @@ def main():
     for _ in range(max(args.warmup - 3, 1)):
         step_fn(opt)
 
+    if args.graph is None:
+        args.graph = 1 if world == 1 else 0  # RCCL capture untested at N>1
     # Capture the whole training step (fwd+bwd+allreduce+optimizer) into
     # one hipGraph: ~875 kernel launches/step collapse into one replay,
     # recovering the launch-gap idle time (measured 14% at bf16 b64).

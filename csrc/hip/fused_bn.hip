@@ -25,6 +25,9 @@
 namespace {
 
 constexpr int BLOCK = 256;
+// Cross-block reduction uses NSHADOW interleaved accumulator copies to cut
+// same-address atomic contention by 8x; bn_finalize / bn_fold sum them.
+constexpr int NSHADOW = 8;
 
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
 typedef __attribute__((ext_vector_type(4))) float float4v;
@@ -93,8 +96,9 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
         atomicAdd(&lds[C + g * 8 + k], q[k]);
     }
     __syncthreads();
+    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-        atomicAdd(&sums[i], lds[i]);
+        atomicAdd(&shadow[i], lds[i]);
     }
 }
 
@@ -114,8 +118,13 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
 {
     const int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
-    const float mean = sums[c] / (float)M;
-    const float var = fmaxf(sums[C + c] / (float)M - mean * mean, 0.f);
+    float s0 = 0.f, s1 = 0.f;
+    for (int k = 0; k < NSHADOW; ++k) {
+        s0 += sums[(size_t)k * 2 * C + c];
+        s1 += sums[(size_t)k * 2 * C + C + c];
+    }
+    const float mean = s0 / (float)M;
+    const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
     const float rstd = rsqrtf(var + eps);
     save_mean[c] = mean;
     save_rstd[c] = rstd;
@@ -131,6 +140,8 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
 }
 
 // ---- pass 2: y = [relu]( a*x + b [+ res] ) ----
+// Fixed channel-octet per thread: per-channel params live in registers for
+// the whole row loop; each block streams contiguous 2*BLOCK*8-byte spans.
 template <bool RELU, bool RES>
 __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
                               const unsigned short *__restrict__ res,
@@ -140,25 +151,27 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
                               int C)
 {
     const int gpr = C / 8;
-    const long long total = M * gpr;  // vec8 groups
-    const long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-         j < total; j += stride) {
-        const int g = (int)(j % gpr);
-        const long long base = (j / gpr) * C + (long long)g * 8;
+    const int rows_per_blk = BLOCK / gpr;
+    const int g = threadIdx.x % gpr;
+    const int row_off = threadIdx.x / gpr;
+    if (row_off >= rows_per_blk) return;
+    float ar[8], br[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        ar[k] = a[g * 8 + k];
+        br[k] = b[g * 8 + k];
+    }
+    const long long row_step = (long long)gridDim.x * rows_per_blk;
+    for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+         r < M; r += row_step) {
+        const long long base = r * C + (long long)g * 8;
         const ushort8 v = *(const ushort8 *)(x + base);
         ushort8 rv;
         if (RES) rv = *(const ushort8 *)(res + base);
         ushort8 out;
-        const float4v a0 = *(const float4v *)(a + g * 8);
-        const float4v a1 = *(const float4v *)(a + g * 8 + 4);
-        const float4v b0 = *(const float4v *)(b + g * 8);
-        const float4v b1 = *(const float4v *)(b + g * 8 + 4);
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
-            const float ak = k < 4 ? a0[k] : a1[k - 4];
-            const float bk = k < 4 ? b0[k] : b1[k - 4];
-            float u = fmaf(ak, b2f(v[k]), bk);
+            float u = fmaf(ar[k], b2f(v[k]), br[k]);
             if (RES) u += b2f(rv[k]);
             if (RELU) u = fmaxf(u, 0.f);
             out[k] = f2b(u);
@@ -191,6 +204,15 @@ __global__ void bn_bwd_reduce_kernel(
     float s1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     float s2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     if (row_off < rows_per_blk) {
+        float ar[8], br[8], mr[8], rr[8];
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            const int c = g * 8 + k;
+            ar[k] = a[c];
+            br[k] = b[c];
+            mr[k] = mean[c];
+            rr[k] = rstd[c];
+        }
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
              r < M; r += row_step) {
@@ -201,15 +223,14 @@ __global__ void bn_bwd_reduce_kernel(
             if (RES) rv = *(const ushort8 *)(res + base);
 #pragma unroll
             for (int k = 0; k < 8; ++k) {
-                const int c = g * 8 + k;
                 const float xf = b2f(xv[k]);
                 float d = b2f(dv[k]);
                 if (RELU) {
-                    float u = fmaf(a[c], xf, b[c]);
+                    float u = fmaf(ar[k], xf, br[k]);
                     if (RES) u += b2f(rv[k]);
                     d = u > 0.f ? d : 0.f;
                 }
-                const float xh = (xf - mean[c]) * rstd[c];
+                const float xh = (xf - mr[k]) * rr[k];
                 s1[k] += d;
                 s2[k] += d * xh;
             }
@@ -221,9 +242,20 @@ __global__ void bn_bwd_reduce_kernel(
         atomicAdd(&lds[C + g * 8 + k], s2[k]);
     }
     __syncthreads();
+    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-        atomicAdd(&sums[i], lds[i]);
+        atomicAdd(&shadow[i], lds[i]);
     }
+}
+
+// Fold the NSHADOW accumulator copies into copy 0.
+__global__ void bn_fold_kernel(float *__restrict__ sums, int C)
+{
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= 2 * C) return;
+    float acc = sums[i];
+    for (int k = 1; k < NSHADOW; ++k) acc += sums[(size_t)k * 2 * C + i];
+    sums[i] = acc;
 }
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
@@ -239,13 +271,26 @@ __global__ void bn_bwd_dx_kernel(
     unsigned short *__restrict__ dx, unsigned short *__restrict__ dres)
 {
     const int gpr = C / 8;
-    const long long total = M * gpr;
-    const long long stride = (long long)gridDim.x * blockDim.x;
+    const int rows_per_blk = BLOCK / gpr;
+    const int g = threadIdx.x % gpr;
+    const int row_off = threadIdx.x / gpr;
+    if (row_off >= rows_per_blk) return;
     const float invM = 1.f / (float)M;
-    for (long long j = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-         j < total; j += stride) {
-        const int g = (int)(j % gpr);
-        const long long base = (j / gpr) * C + (long long)g * 8;
+    float ar[8], br[8], mr[8], rr[8], t1[8], t2[8];
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        const int c = g * 8 + k;
+        ar[k] = a[c];
+        br[k] = b[c];
+        mr[k] = mean[c];
+        rr[k] = rstd[c];
+        t1[k] = sums[c] * invM;      // mean of dy_m
+        t2[k] = sums[C + c] * invM;  // mean of dy_m * xhat
+    }
+    const long long row_step = (long long)gridDim.x * rows_per_blk;
+    for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+         r < M; r += row_step) {
+        const long long base = r * C + (long long)g * 8;
         const ushort8 dv = *(const ushort8 *)(dy + base);
         const ushort8 xv = *(const ushort8 *)(x + base);
         ushort8 rv;
@@ -253,19 +298,17 @@ __global__ void bn_bwd_dx_kernel(
         ushort8 dxo, dro;
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
-            const int c = g * 8 + k;
             const float xf = b2f(xv[k]);
             float d = b2f(dv[k]);
             if (RELU) {
-                float u = fmaf(a[c], xf, b[c]);
+                float u = fmaf(ar[k], xf, br[k]);
                 if (RES) u += b2f(rv[k]);
                 d = u > 0.f ? d : 0.f;
             }
             if (RES) dro[k] = f2b(d);
-            const float xh = (xf - mean[c]) * rstd[c];
-            const float t =
-                d - sums[c] * invM - xh * (sums[C + c] * invM);
-            dxo[k] = f2b(a[c] * t);
+            const float xh = (xf - mr[k]) * rr[k];
+            const float t = d - t1[k] - xh * t2[k];
+            dxo[k] = f2b(ar[k] * t);
         }
         *(ushort8 *)(dx + base) = dxo;
         if (RES) *(ushort8 *)(dres + base) = dro;
@@ -287,6 +330,9 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
     long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    // >= 64 KiB per block so the atomic epilogue stays cold on small layers
+    const long long by_bytes = (M * C * 2 + 65535) / 65536;
+    if (blocks > by_bytes) blocks = by_bytes;
     if (blocks > 2048) blocks = 2048;
     if (blocks < 1) blocks = 1;
     hipLaunchKernelGGL(bn_stats_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
@@ -314,8 +360,12 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
                      const void *b, long long M, int C, int relu,
                      void *stream)
 {
-    const long long total = M * (C / 8);
-    const dim3 grid(grid_for(total)), block(BLOCK);
+    if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
+    const int rows_per_blk = BLOCK / (C / 8);
+    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
 #define CASE(R, E)                                                          \
     hipLaunchKernelGGL((bn_fwd_kernel<R, E>), grid, block, 0, s,            \
@@ -339,6 +389,8 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x, const void *res,
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
     long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    const long long by_bytes = (M * C * 4 + 65535) / 65536;  // 2 streams
+    if (blocks > by_bytes) blocks = by_bytes;
     if (blocks > 2048) blocks = 2048;
     if (blocks < 1) blocks = 1;
     const dim3 grid((uint32_t)blocks), block(BLOCK);
@@ -359,13 +411,24 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x, const void *res,
     return hipGetLastError();
 }
 
+hipError_t kf_bn_fold(void *sums, int C, void *stream)
+{
+    hipLaunchKernelGGL(bn_fold_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (float *)sums, C);
+    return hipGetLastError();
+}
+
 hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *res,
                         const void *a, const void *b, const void *mean,
                         const void *rstd, const void *sums, long long M,
                         int C, int relu, void *dx, void *dres, void *stream)
 {
-    const long long total = M * (C / 8);
-    const dim3 grid(grid_for(total)), block(BLOCK);
+    if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
+    const int rows_per_blk = BLOCK / (C / 8);
+    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
 #define CASE(R, E)                                                          \
     hipLaunchKernelGGL((bn_bwd_dx_kernel<R, E>), grid, block, 0, s,         \

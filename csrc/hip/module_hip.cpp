@@ -27,6 +27,7 @@ hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, const void *,
                             const void *, long long, int, int, void *,
                             void *);
+hipError_t kf_bn_fold(void *, int, void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
                         const void *, const void *, long long, int, int,
@@ -232,6 +233,9 @@ PYBIND11_MODULE(_hip, m)
                                      (void *)stream),
                     "kf_bn_bwd_reduce");
           });
+    m.def("bn_fold", [](uintptr_t sums, int C, uintptr_t stream) {
+        check(kf_bn_fold((void *)sums, C, (void *)stream), "kf_bn_fold");
+    });
     m.def("bn_bwd_dx",
           [](uintptr_t dy, uintptr_t x, uintptr_t res, uintptr_t a,
              uintptr_t b, uintptr_t mean, uintptr_t rstd, uintptr_t sums,

@@ -37,7 +37,7 @@ class _FusedBNFunction(torch.autograd.Function):
         s = _stream()
         dev = x.device
         if training:
-            sums = torch.zeros(2 * C, dtype=torch.float32, device=dev)
+            sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
             _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)
             save_mean = torch.empty(C, dtype=torch.float32, device=dev)
             save_rstd = torch.empty(C, dtype=torch.float32, device=dev)
@@ -73,12 +73,13 @@ class _FusedBNFunction(torch.autograd.Function):
         s = _stream()
         dy = dy.contiguous(memory_format=torch.channels_last)
         dev = x.device
-        sums = torch.zeros(2 * C, dtype=torch.float32, device=dev)
+        sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
         res_ptr = residual.data_ptr() if ctx.has_res else 0
         _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), res_ptr,
                            a.data_ptr(), b.data_ptr(),
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
                            C, ctx.relu, sums.data_ptr(), s)
+        _hip.bn_fold(sums.data_ptr(), C, s)  # fold shadow accumulators
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), res_ptr, a.data_ptr(),
@@ -86,8 +87,8 @@ class _FusedBNFunction(torch.autograd.Function):
                        save_rstd.data_ptr(), sums.data_ptr(), M, C,
                        ctx.relu, dx.data_ptr(),
                        dres.data_ptr() if dres is not None else 0, s)
-        db = sums[:C].clone()          # db = sum(dy_m)
-        dw = sums[C:].clone()          # dw = sum(dy_m * xhat)
+        db = sums[:C]               # db = sum(dy_m)
+        dw = sums[C:2 * C]          # dw = sum(dy_m * xhat)
         return (dx, dres, dw, db, None, None, None, None, None, None)
 
 
