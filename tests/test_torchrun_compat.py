@@ -1,0 +1,79 @@
+"""The driver launches bench.py via torch.distributed.run — verify the
+RANK/WORLD_SIZE env-synthesis path (no kungfu-run) end to end on CPU."""
+import os
+import re
+import subprocess
+import sys
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def test_bench_slp_under_torchrun_env(port_block):
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({
+            "PYTHONPATH": ROOT,
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "2",
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port_block + 40),
+            "KUNGFU_PORT_BASE": str(port_block),
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, "bench.py", "--model", "slp", "--steps", "4",
+             "--warmup", "1", "--dtype", "fp32"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=180)
+        assert p.returncode == 0, out + err
+        outs.append(out)
+    # exactly one rank (rank 0) prints the JSON line with n_gpus=2
+    jsons = [ln for out in outs for ln in out.splitlines()
+             if ln.startswith("{")]
+    assert len(jsons) == 1, outs
+    assert '"n_gpus": 2' in jsons[0]
+
+
+def test_elastic_trainer_class(port_block):
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    script = r"""
+import torch
+import kungfu_amd as kf
+from kungfu_amd.models import SLP
+from kungfu_amd.optimizers import SynchronousSGDOptimizer
+from kungfu_amd.parallel.elastic import ElasticTrainer
+
+kf.init(with_torch=False)
+torch.manual_seed(0)
+trainer = ElasticTrainer(
+    SLP(in_features=8, classes=2),
+    lambda m: SynchronousSGDOptimizer(torch.optim.SGD(m.parameters(),
+                                                      lr=0.1)),
+    schedule="2:3")
+while trainer.step < 5:
+    x = torch.randn(4, 1, 2, 4)
+    y = torch.randint(0, 2, (4,))
+    trainer.optimizer.zero_grad()
+    torch.nn.functional.cross_entropy(trainer.model(x), y).backward()
+    trainer.optimizer.step()
+    if not trainer.after_step():
+        break
+if not trainer.detached:
+    print("ET-DONE size=%d step=%d" % (kf.size(), trainer.step))
+kf.finalize()
+"""
+    r = subprocess.run(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "2", "-w",
+         "-port", str(port_block), "-port-range", str(port_block + 2),
+         "-builtin-config-port", str(port_block + 1),
+         sys.executable, "-c", script],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=240)
+    out = re.sub(r"\x1b\[[0-9;]*m", "", r.stdout)
+    assert r.returncode == 0, out + r.stderr
+    done = [ln for ln in out.splitlines() if "ET-DONE" in ln]
+    assert len(done) == 3 and all("size=3 step=5" in ln for ln in done), out
