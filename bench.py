@@ -42,6 +42,8 @@ def parse_args():
     p.add_argument("--graph", type=int, default=None,
                    help="capture the step into a hipGraph and replay "
                         "(default: on for single-GPU runs)")
+    p.add_argument("--fused-opt", type=int, default=1,
+                   help="fused SGD-momentum HIP kernel over flat buckets")
     p.add_argument("--fused-bn", type=int, default=1,
                    help="resnet50: use the gfx950 fused BN(+res+ReLU) "
                         "kernels instead of autocast fp32 BN")
@@ -128,9 +130,12 @@ def wrap_optimizer(args, model):
 
     inner = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
     if args.optimizer == "sync":
+        import torch as _t
+
         return kfo.SynchronousSGDOptimizer(
             inner, bucket_bytes=args.bucket_mb << 20,
-            overlap=not args.no_overlap)
+            overlap=not args.no_overlap,
+            fused_step=bool(args.fused_opt) and _t.cuda.is_available())
     if args.optimizer == "sma":
         return kfo.SynchronousAveragingOptimizer(inner)
     if args.optimizer == "pair":
@@ -153,6 +158,11 @@ def max_over_ranks(value):
 
 def main():
     args = parse_args()
+    # ship tuned MIOpen find-db results with the repo (benchmarks/tune)
+    _tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                          "tuned", "miopen")
+    if os.path.isdir(_tuned):
+        os.environ.setdefault("MIOPEN_USER_DB_PATH", _tuned)
     torch.backends.cudnn.benchmark = True  # MIOpen find once per shape
     import kungfu_amd as kf
 

@@ -84,3 +84,33 @@ def test_sma_pair_gpu_single():
         torch.nn.functional.cross_entropy(m(x).float(), y).backward()
         opt.step()
     torch.cuda.synchronize()
+
+
+def test_fused_sgd_step_matches_torch():
+    import kungfu_amd as kf
+    from kungfu_amd.optimizers import SynchronousSGDOptimizer
+
+    kf.init()
+    torch.manual_seed(5)
+    lin1 = torch.nn.Sequential(torch.nn.Linear(64, 64),
+                               torch.nn.Linear(64, 8)).to("cuda")
+    lin2 = torch.nn.Sequential(torch.nn.Linear(64, 64),
+                               torch.nn.Linear(64, 8)).to("cuda")
+    lin2.load_state_dict(lin1.state_dict())
+    opt1 = SynchronousSGDOptimizer(
+        torch.optim.SGD(lin1.parameters(), lr=0.1, momentum=0.9,
+                        weight_decay=1e-4), fused_step=True)
+    opt2 = torch.optim.SGD(lin2.parameters(), lr=0.1, momentum=0.9,
+                           weight_decay=1e-4)
+    for i in range(3):
+        x = torch.randn(16, 64, device="cuda")
+        opt1.zero_grad()
+        lin1(x).pow(2).mean().backward()
+        opt1.step()
+        opt2.zero_grad()
+        lin2(x).pow(2).mean().backward()
+        opt2.step()
+    torch.cuda.synchronize()
+    for p1, p2 in zip(lin1.parameters(), lin2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-5, rtol=1e-5), \
+            (p1 - p2).abs().max()
