@@ -43,7 +43,7 @@ def _alias_view(flat_slice, p):
 
 class _Bucket:
     __slots__ = ("index", "params", "flat", "numel", "ready", "work",
-                 "launched")
+                 "launched", "param_flat", "momentum")
 
     def __init__(self, index):
         self.index = index
