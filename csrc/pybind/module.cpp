@@ -8,10 +8,12 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <chrono>
 #include <cstring>
 #include <deque>
 #include <memory>
 #include <mutex>
+#include <tuple>
 
 #include "../core/graph.hpp"
 #include "../core/plan.hpp"
@@ -31,20 +33,68 @@ Peer &peer()
     return *g_peer;
 }
 
-// RAII stall-detection guard around blocking collective calls
+// Runtime tracing (reference: stdtracer TRACE_SCOPE, KUNGFU_ENABLE_TRACE;
+// include/kungfu/utils/trace.hpp): env-gated event ring readable from
+// python as chrome-trace-able tuples.
+struct Tracer {
+    bool enabled = false;
+    std::mutex mu;
+    std::deque<std::tuple<std::string, double, double>> events;  // us
+    std::chrono::steady_clock::time_point t0 =
+        std::chrono::steady_clock::now();
+
+    static Tracer &get()
+    {
+        static Tracer t;
+        static bool init = [] {
+            const char *e = std::getenv("KUNGFU_ENABLE_TRACE");
+            get_unsafe().enabled = e && *e && std::string(e) != "0";
+            return true;
+        }();
+        (void)init;
+        return get_unsafe();
+    }
+    static Tracer &get_unsafe()
+    {
+        static Tracer t;
+        return t;
+    }
+    void record(const char *what, double start_us, double dur_us)
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        if (events.size() > 100000) events.pop_front();
+        events.emplace_back(what, start_us, dur_us);
+    }
+};
+
+// RAII guard around blocking collective calls: stall detection + tracing
 struct StallGuard {
     StallDetector *d = nullptr;
     uint64_t id = 0;
-    explicit StallGuard(const char *what)
+    const char *what_;
+    std::chrono::steady_clock::time_point start_;
+    explicit StallGuard(const char *what) : what_(what)
     {
         if (g_peer && g_peer->stall_detector()) {
             d = g_peer->stall_detector();
             id = d->enter(what);
         }
+        if (Tracer::get().enabled)
+            start_ = std::chrono::steady_clock::now();
     }
     ~StallGuard()
     {
         if (d) d->leave(id);
+        auto &t = Tracer::get();
+        if (t.enabled) {
+            auto now = std::chrono::steady_clock::now();
+            t.record(
+                what_,
+                std::chrono::duration<double, std::micro>(start_ - t.t0)
+                    .count(),
+                std::chrono::duration<double, std::micro>(now - start_)
+                    .count());
+        }
     }
 };
 
@@ -194,6 +244,7 @@ PYBIND11_MODULE(_core, m)
              const std::string &name, int root) {
               auto w = make_ws(s, r, count, dt, 0, name);
               py::gil_scoped_release rel;
+              StallGuard sg("broadcast");
               peer().session().broadcast(w, root);
           });
     m.def("all_gather",
@@ -300,6 +351,17 @@ PYBIND11_MODULE(_core, m)
     m.def("check_interference", [](double ratio) {
         return peer().session().check_interference(ratio);
     });
+    m.def("trace_events", [] {
+        auto &t = Tracer::get();
+        std::lock_guard<std::mutex> lk(t.mu);
+        py::list out;
+        for (auto &e : t.events) {
+            out.append(py::make_tuple(std::get<0>(e), std::get<1>(e),
+                                      std::get<2>(e)));
+        }
+        return out;
+    });
+    m.def("trace_enabled", [] { return Tracer::get().enabled; });
     m.def("egress_bytes", [] {
         auto eg = peer().egress_bytes();
         py::dict d;

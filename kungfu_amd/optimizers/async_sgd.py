@@ -26,10 +26,14 @@ _MODEL_KEY = "model"
 
 class PairAveragingOptimizer(KungFuOptimizer):
     def __init__(self, optimizer, fuse_requests=True, prefetch=False,
-                 name=_MODEL_KEY):
+                 name=_MODEL_KEY, peer_selection="random"):
+        """peer_selection: 'random' (AD-PSGD gossip) or 'roundrobin'
+        (reference GetNeighbour/RoundRobin ops, ops/cpu/topology.cpp)."""
         super().__init__(optimizer)
         self.name = name
         self.prefetch = prefetch
+        self.peer_selection = peer_selection
+        self._rr_step = 0
         self._group = FlatParamGroup(self._params())
         self._host_buf = torch.empty(self._group.numel,
                                      dtype=self._group.dtype, device="cpu")
@@ -54,6 +58,10 @@ class PairAveragingOptimizer(KungFuOptimizer):
         np_, r = _core.size(), _core.rank()
         if np_ <= 1:
             return -1
+        if self.peer_selection == "roundrobin":
+            self._rr_step += 1
+            off = 1 + (self._rr_step - 1) % (np_ - 1)
+            return (r + off) % np_
         t = random.randrange(np_ - 1)
         return t if t < r else t + 1
 
