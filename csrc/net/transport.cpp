@@ -176,6 +176,12 @@ void Server::accept_loop(int listen_fd)
             return;
         }
         std::lock_guard<std::mutex> lk(mu_);
+        if (stopping_.load()) {
+            // stop() may already be joining the (moved-out) thread list;
+            // mutating threads_ here would race its iteration
+            ::close(fd);
+            return;
+        }
         threads_.emplace_back([this, fd] { handle_conn(fd); });
     }
 }
@@ -242,14 +248,16 @@ void Server::stop()
         ::close(unix_fd_);
         ::unlink(unix_sock_path(self_.port).c_str());
     }
+    std::vector<std::thread> ts;
     {
         std::lock_guard<std::mutex> lk(mu_);
         for (auto &c : conns_) c->close_fd();
+        ts = std::move(threads_);  // join outside the lock, accept_loop
+        threads_.clear();          // can no longer add (stopping_ set)
     }
-    for (auto &t : threads_) {
+    for (auto &t : ts) {
         if (t.joinable()) t.join();
     }
-    threads_.clear();
 }
 
 // ---------- Client ----------

@@ -28,8 +28,11 @@ def run_worker(fn, i, np, base, strategy, q, extra=None):
         q.put((i, "err", "%s\n%s" % (e, traceback.format_exc())))
 
 
-def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None):
-    """Run fn(rank, np) in np processes; returns list of results."""
+def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None,
+                  _retry=True):
+    """Run fn(rank, np) in np processes; returns list of results.
+    Retries once on a port-bind collision (stray listener on the random
+    test port block)."""
     import multiprocessing as mp
 
     ctx = mp.get_context("spawn")
@@ -51,6 +54,12 @@ def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None):
             except queue_mod.Empty:
                 raise AssertionError("cluster timed out; results so far: %r"
                                      % (results,))
+            if (status == "err" and _retry
+                    and "Address already in use" in str(payload)):
+                for p in procs:
+                    p.terminate()
+                return spawn_cluster(fn, np, base + 8192, strategy,
+                                     timeout, extra, _retry=False)
             assert status == "ok", "rank %d failed: %s" % (i, payload)
             results[i] = payload
     finally:
