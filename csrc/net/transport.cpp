@@ -144,7 +144,12 @@ void Server::start(FrameHandler handler, std::function<bool(uint32_t)> token_ok)
                                  std::strerror(errno));
     if (::listen(tcp_fd_, 128) != 0)
         throw std::runtime_error("listen failed");
-    threads_.emplace_back([this] { accept_loop(tcp_fd_); });
+    {
+        // threads_ is shared with the accept loops (they add handler
+        // threads under mu_): every mutation must hold the lock
+        std::lock_guard<std::mutex> lk(mu_);
+        threads_.emplace_back([this] { accept_loop(tcp_fd_); });
+    }
 
     if (use_unix_) {
         unix_fd_ = ::socket(AF_UNIX, SOCK_STREAM, 0);
@@ -157,6 +162,7 @@ void Server::start(FrameHandler handler, std::function<bool(uint32_t)> token_ok)
                           path.c_str());
             if (::bind(unix_fd_, (sockaddr *)&ua, sizeof(ua)) == 0 &&
                 ::listen(unix_fd_, 128) == 0) {
+                std::lock_guard<std::mutex> lk(mu_);
                 threads_.emplace_back([this] { accept_loop(unix_fd_); });
             } else {
                 ::close(unix_fd_);

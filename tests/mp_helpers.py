@@ -19,6 +19,11 @@ def set_env(i, np, base, strategy="AUTO", extra=None):
 
 def run_worker(fn, i, np, base, strategy, q, extra=None):
     try:
+        import faulthandler
+        import signal
+
+        faulthandler.enable()
+        faulthandler.register(signal.SIGUSR1, all_threads=True)
         set_env(i, np, base, strategy, extra)
         result = fn(i, np)
         q.put((i, "ok", result))
@@ -52,6 +57,18 @@ def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None,
             try:
                 i, status, payload = q.get(timeout=timeout)
             except queue_mod.Empty:
+                import signal
+                import time as time_mod
+
+                for p in procs:  # dump stuck children's stacks to stderr
+                    if p.is_alive():
+                        try:
+                            import os as os_mod
+
+                            os_mod.kill(p.pid, signal.SIGUSR1)
+                        except OSError:
+                            pass
+                time_mod.sleep(2)
                 raise AssertionError("cluster timed out; results so far: %r"
                                      % (results,))
             if (status == "err" and _retry
