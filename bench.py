@@ -29,7 +29,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-size", type=int, default=64)
     p.add_argument("--model", default="resnet50",
-                   choices=["resnet50", "bert", "slp"])
+                   choices=["resnet50", "vgg16", "bert", "slp"])
     p.add_argument("--optimizer", default="sync",
                    choices=["sync", "sma", "pair", "gns"])
     p.add_argument("--seq-len", type=int, default=128)
@@ -71,6 +71,27 @@ def build_model_and_data(args, device, dtype, amp):
         y = torch.randint(0, 1000, (args.batch_size,), device=device)
         model = model.to(device=device, dtype=dtype)
         x = x.to(device=device, dtype=dtype)
+        if args.channels_last and device.type == "cuda":
+            model = model.to(memory_format=torch.channels_last)
+            x = x.contiguous(memory_format=torch.channels_last)
+
+        def step_fn(opt):
+            opt.zero_grad()
+            with amp_ctx():
+                out = model(x)
+            loss = torch.nn.functional.cross_entropy(out.float(), y)
+            loss.backward()
+            opt.step()
+            return loss
+
+        return model, step_fn, args.batch_size
+    if args.model == "vgg16":
+        from kungfu_amd.models import vgg16
+
+        model = vgg16().to(device=device, dtype=dtype)
+        x = torch.randn(args.batch_size, 3, 224, 224, device=device,
+                        dtype=dtype)
+        y = torch.randint(0, 1000, (args.batch_size,), device=device)
         if args.channels_last and device.type == "cuda":
             model = model.to(memory_format=torch.channels_last)
             x = x.contiguous(memory_format=torch.channels_last)

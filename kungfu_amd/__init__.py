@@ -68,35 +68,41 @@ def _maybe_init_torch_dist():
         return
     import torch
 
-    if not torch.cuda.is_available():
-        return  # CPU plumbing mode uses the C++ engine only
+    backend = os.environ.get("KUNGFU_TORCH_BACKEND")  # tests: "gloo"
+    if backend is None:
+        if not torch.cuda.is_available():
+            return  # CPU plumbing mode uses the C++ engine only
+        backend = "nccl"  # RCCL on ROCm
     import torch.distributed as dist
 
     if dist.is_initialized():
         _torch_dist = True
         return
     rank, world = _core.rank(), _core.size()
-    if torch.cuda.device_count() >= world:
-        # single launcher-less process group (e.g. torchrun on one node)
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    if torch.cuda.is_available():
+        if torch.cuda.device_count() >= world:
+            # launcher-less run (e.g. torchrun on one node)
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        else:
+            torch.cuda.set_device(0)  # kungfu-run pins HIP_VISIBLE_DEVICES
+    if os.environ.get("MASTER_ADDR") and os.environ.get("MASTER_PORT"):
+        # torchrun path: attach to the launcher's store (env://) — creating
+        # our own TCPStore on MASTER_PORT would collide with torchrun's
+        dist.init_process_group(backend=backend, rank=rank,
+                                world_size=world)
     else:
-        torch.cuda.set_device(0)  # kungfu-run pins HIP_VISIBLE_DEVICES
-    addr = os.environ.get("MASTER_ADDR")
-    port = os.environ.get("MASTER_PORT")
-    if not addr:
+        # kungfu-run path: rank 0 hosts a fresh store on a derived port
+        # (re-derived per cluster version so elastic re-inits don't clash)
         first = os.environ["KUNGFU_INIT_PEERS"].split(",")[0]
         addr = first.rsplit(":", 1)[0]
-        port = None
-    if not port:
-        first_port = int(
-            os.environ["KUNGFU_INIT_PEERS"].split(",")[0].rsplit(":", 1)[1])
-        port = str(first_port + 1711 + int(_core.cluster_version()))
-    dist.init_process_group(
-        backend="nccl",
-        init_method="tcp://%s:%s" % (addr, port),
-        rank=rank,
-        world_size=world,
-    )
+        port = int(first.rsplit(":", 1)[1]) + 1711 + \
+            int(_core.cluster_version())
+        dist.init_process_group(
+            backend=backend,
+            init_method="tcp://%s:%d" % (addr, port),
+            rank=rank,
+            world_size=world,
+        )
     _torch_dist = True
 
 

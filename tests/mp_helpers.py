@@ -13,6 +13,9 @@ def set_env(i, np, base, strategy="AUTO", extra=None):
     os.environ["KUNGFU_INIT_PEERS"] = peers
     os.environ["KUNGFU_ALLREDUCE_STRATEGY"] = strategy
     os.environ.pop("KUNGFU_CONFIG_SERVER", None)
+    for k in ("MASTER_ADDR", "MASTER_PORT", "KUNGFU_TORCH_BACKEND",
+              "RANK", "WORLD_SIZE", "LOCAL_RANK"):
+        os.environ.pop(k, None)
     for k, v in (extra or {}).items():
         os.environ[k] = v
 
@@ -300,3 +303,17 @@ def p2p_bigpull_body(rank, np):
     kf.barrier()
     kf.finalize()
     return blob.nbytes / dt / 1e9  # GB/s
+
+
+def dist_gloo_body(rank, np):
+    import torch
+    import torch.distributed as dist
+    import kungfu_amd as kf
+
+    kf.init()  # KUNGFU_TORCH_BACKEND=gloo set via extra env
+    assert dist.is_initialized(), "process group did not come up"
+    t = torch.ones(10) * (rank + 1)
+    dist.all_reduce(t)
+    out = float(t[0])
+    kf.finalize()
+    return out

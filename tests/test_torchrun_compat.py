@@ -77,3 +77,26 @@ kf.finalize()
     assert r.returncode == 0, out + r.stderr
     done = [ln for ln in out.splitlines() if "ET-DONE" in ln]
     assert len(done) == 3 and all("size=3 step=5" in ln for ln in done), out
+
+
+def test_process_group_env_master(port_block):
+    """torchrun-style: MASTER_ADDR/PORT set -> env:// attach (gloo)."""
+    from mp_helpers import dist_gloo_body, spawn_cluster
+
+    extra = {
+        "KUNGFU_TORCH_BACKEND": "gloo",
+        "MASTER_ADDR": "127.0.0.1",
+        "MASTER_PORT": str(port_block + 50),
+    }
+    res = spawn_cluster(dist_gloo_body, 2, port_block, extra=extra)
+    assert res == [3.0, 3.0]
+
+
+def test_process_group_derived_store(port_block):
+    """kungfu-run style: no MASTER env -> rank0 hosts a derived-port
+    store."""
+    from mp_helpers import dist_gloo_body, spawn_cluster
+
+    extra = {"KUNGFU_TORCH_BACKEND": "gloo"}
+    res = spawn_cluster(dist_gloo_body, 2, port_block, extra=extra)
+    assert res == [3.0, 3.0]
