@@ -29,7 +29,8 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=10)
     p.add_argument("--batch-size", type=int, default=64)
     p.add_argument("--model", default="resnet50",
-                   choices=["resnet50", "vgg16", "bert", "slp"])
+                   choices=["resnet50", "vgg16", "inception-v3", "bert",
+                            "slp"])
     p.add_argument("--optimizer", default="sync",
                    choices=["sync", "sma", "pair", "gns"])
     p.add_argument("--seq-len", type=int, default=128)
@@ -85,11 +86,13 @@ def build_model_and_data(args, device, dtype, amp):
             return loss
 
         return model, step_fn, args.batch_size
-    if args.model == "vgg16":
-        from kungfu_amd.models import vgg16
+    if args.model in ("vgg16", "inception-v3"):
+        from kungfu_amd.models import inception_v3, vgg16
 
-        model = vgg16().to(device=device, dtype=dtype)
-        x = torch.randn(args.batch_size, 3, 224, 224, device=device,
+        size = 224 if args.model == "vgg16" else 299
+        model = (vgg16() if args.model == "vgg16" else
+                 inception_v3()).to(device=device, dtype=dtype)
+        x = torch.randn(args.batch_size, 3, size, size, device=device,
                         dtype=dtype)
         y = torch.randint(0, 1000, (args.batch_size,), device=device)
         if args.channels_last and device.type == "cuda":

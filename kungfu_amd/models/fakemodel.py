@@ -23,6 +23,7 @@ def model_sizes(name):
         "vgg16-imagenet": lambda: models.vgg16(),
         "bert": lambda: models.bert_base(),
         "slp-mnist": lambda: models.SLP(),
+        "inception-v3": lambda: models.inception_v3(),
     }
     if name not in builders:
         raise KeyError("unknown fake model %r (have %s)" %

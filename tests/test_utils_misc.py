@@ -89,3 +89,30 @@ def test_round_robin_peer_selection():
     finally:
         mod._core = __import__("kungfu_amd")._core
     assert picks == [2, 3, 0, 2, 3, 0]  # cycles over all other ranks
+
+
+def test_inception_v3_shapes():
+    import torch
+    from kungfu_amd.models import inception_v3
+
+    m = inception_v3()
+    n = sum(p.numel() for p in m.parameters())
+    assert 23_500_000 < n < 24_500_000  # canonical ~23.8M
+    with torch.no_grad():
+        assert m(torch.randn(1, 3, 299, 299)).shape == (1, 1000)
+
+
+def test_prim_mst():
+    from kungfu_amd import _core
+
+    # 4 nodes; cheapest tree = 0-1 (1), 1-2 (1), 0-3 (2)
+    inf = 100.0
+    w = [0, 1, inf, 2,
+         1, 0, 1, inf,
+         inf, 1, 0, inf,
+         2, inf, inf, 0]
+    parent = _core.prim_mst([float(v) for v in w], 4)
+    assert parent[0] == 0
+    assert parent[1] == 0
+    assert parent[2] == 1
+    assert parent[3] == 0
