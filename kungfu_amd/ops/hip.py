@@ -31,12 +31,21 @@ def _stream():
     return torch.cuda.current_stream().cuda_stream
 
 
+def _dense(t):
+    # dense storage in SOME layout (row-major or channels_last): raw
+    # pointer kernels only need the numel elements to be one dense block
+    if t.is_contiguous():
+        return True
+    return t.dim() == 4 and t.is_contiguous(
+        memory_format=torch.channels_last)
+
+
 def _check_cuda(*tensors):
     for t in tensors:
         if not t.is_cuda:
             raise ValueError("HIP kernel requires CUDA(HIP) tensors")
-        if not t.is_contiguous():
-            raise ValueError("HIP kernel requires contiguous tensors")
+        if not _dense(t):
+            raise ValueError("HIP kernel requires dense tensors")
 
 
 class FusionPlan:

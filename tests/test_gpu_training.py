@@ -114,3 +114,26 @@ def test_fused_sgd_step_matches_torch():
     for p1, p2 in zip(lin1.parameters(), lin2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-5, rtol=1e-5), \
             (p1 - p2).abs().max()
+
+
+def test_sma_pair_with_channels_last_model():
+    """SMA/Pair fuse channels_last conv weights through the raw-storage
+    pack kernels (regression: dense-but-not-row-major layouts)."""
+    import kungfu_amd as kf
+    from kungfu_amd.models import resnet50
+    from kungfu_amd.optimizers import (PairAveragingOptimizer,
+                                       SynchronousAveragingOptimizer)
+
+    kf.init()
+    for cls in (SynchronousAveragingOptimizer, PairAveragingOptimizer):
+        m = resnet50().to("cuda").to(memory_format=torch.channels_last)
+        opt = cls(torch.optim.SGD(m.parameters(), lr=0.01))
+        x = torch.randn(2, 3, 64, 64, device="cuda").contiguous(
+            memory_format=torch.channels_last)
+        y = torch.randint(0, 1000, (2,), device="cuda")
+        opt.zero_grad()
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            out = m(x)
+        torch.nn.functional.cross_entropy(out.float(), y).backward()
+        opt.step()
+    torch.cuda.synchronize()
