@@ -55,6 +55,13 @@ void set_nodelay(int fd)
     ::setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof(one));
 }
 
+void set_bufsizes(int fd)
+{
+    int sz = 4 << 20;  // deep buffers: the collective engine streams MiB
+    ::setsockopt(fd, SOL_SOCKET, SO_SNDBUF, &sz, sizeof(sz));
+    ::setsockopt(fd, SOL_SOCKET, SO_RCVBUF, &sz, sizeof(sz));
+}
+
 }  // namespace
 
 // ---------- Conn ----------
@@ -195,6 +202,7 @@ void Server::accept_loop(int listen_fd)
 void Server::handle_conn(int fd)
 {
     set_nodelay(fd);
+    set_bufsizes(fd);
     // handshake: magic, type, src ip, src port, token
     uint8_t hs_buf[4 + 1 + 4 + 2 + 4];
     if (!read_all(fd, hs_buf, sizeof(hs_buf))) {
@@ -291,7 +299,10 @@ std::shared_ptr<Conn> Client::get_conn(const PeerID &remote, ConnType type,
                 ua.sun_family = AF_UNIX;
                 std::snprintf(ua.sun_path, sizeof(ua.sun_path), "%s",
                               Server::unix_sock_path(remote.port).c_str());
-                if (::connect(fd, (sockaddr *)&ua, sizeof(ua)) == 0) break;
+                if (::connect(fd, (sockaddr *)&ua, sizeof(ua)) == 0) {
+                    set_bufsizes(fd);
+                    break;
+                }
                 ::close(fd);
                 fd = -1;
             }
@@ -304,6 +315,7 @@ std::shared_ptr<Conn> Client::get_conn(const PeerID &remote, ConnType type,
             addr.sin_port = htons(remote.port);
             if (::connect(fd, (sockaddr *)&addr, sizeof(addr)) == 0) {
                 set_nodelay(fd);
+                set_bufsizes(fd);
                 break;
             }
             ::close(fd);
