@@ -163,7 +163,10 @@ def wrap_optimizer(args, model):
     if args.optimizer == "sma":
         return kfo.SynchronousAveragingOptimizer(inner)
     if args.optimizer == "pair":
-        return kfo.PairAveragingOptimizer(inner)
+        import torch as _t
+
+        return kfo.PairAveragingOptimizer(
+            inner, exchange="rccl" if _t.cuda.is_available() else "store")
     return kfo.MonitorGradientNoiseScaleOptimizer(
         inner, device_batch_size=args.batch_size)
 

@@ -24,15 +24,40 @@ from kungfu_amd.parallel.fusion import FlatParamGroup
 _MODEL_KEY = "model"
 
 
+def tournament_partner(rank, step, n):
+    """Symmetric pairing for round `step` (circle-method round-robin):
+    every rank's partner has this rank as its own partner, and over n-1
+    (n even) rounds each rank meets every other exactly once. Returns -1
+    when this rank sits out (odd n). Used by the RCCL/xGMI exchange mode
+    where both sides of a pair must participate in send/recv."""
+    if n < 2:
+        return -1
+    q = n if n % 2 else n - 1  # odd modulus; with even n, rank n-1 floats
+    m = step % q
+    if n % 2 == 0 and rank == n - 1:
+        return (m * (q + 1) // 2) % q
+    cand = (m - rank) % q
+    if cand == rank:
+        return n - 1 if n % 2 == 0 else -1  # paired with the floater
+    return cand
+
+
 class PairAveragingOptimizer(KungFuOptimizer):
     def __init__(self, optimizer, fuse_requests=True, prefetch=False,
-                 name=_MODEL_KEY, peer_selection="random"):
+                 name=_MODEL_KEY, peer_selection="random",
+                 exchange="store"):
         """peer_selection: 'random' (AD-PSGD gossip) or 'roundrobin'
-        (reference GetNeighbour/RoundRobin ops, ops/cpu/topology.cpp)."""
+        (reference GetNeighbour/RoundRobin ops, ops/cpu/topology.cpp).
+        exchange: 'store' (asymmetric pulls from the P2P blob store, the
+        reference's model) or 'rccl' (symmetric tournament gossip with
+        direct GPU-GPU sendrecv over xGMI — both partners exchange and
+        average on-device; needs the torch.distributed process group)."""
         super().__init__(optimizer)
         self.name = name
         self.prefetch = prefetch
         self.peer_selection = peer_selection
+        self.exchange = exchange
+        self._recv_buf = None
         self._rr_step = 0
         self._group = FlatParamGroup(self._params())
         self._host_buf = torch.empty(self._group.numel,
@@ -82,9 +107,43 @@ class PairAveragingOptimizer(KungFuOptimizer):
         self._prefetch_thread.start()
 
     @torch.no_grad()
+    def _rccl_exchange(self):
+        """xGMI-native gossip: symmetric tournament pairing, direct
+        GPU-to-GPU model sendrecv over RCCL (no host staging), fused
+        on-device averaging. Both partners run the same schedule, so the
+        exchange is deadlock-free by construction."""
+        import torch.distributed as dist
+
+        n, r = _core.size(), _core.rank()
+        partner = tournament_partner(r, self._rr_step, n)
+        self._rr_step += 1
+        if partner < 0:
+            self.optimizer.step()
+            return
+        g = self._group
+        g.pack()  # flat <- v
+        if self._recv_buf is None:
+            self._recv_buf = torch.empty_like(g.flat)
+        ops = [dist.P2POp(dist.isend, g.flat, partner),
+               dist.P2POp(dist.irecv, self._recv_buf, partner)]
+        for work in dist.batch_isend_irecv(ops):
+            work.wait()
+        if g.flat.is_cuda:
+            from kungfu_amd.ops import hip as hip_ops
+
+            hip_ops.avg_inplace(g.flat, self._recv_buf, alpha=0.5)
+        else:
+            g.flat.mul_(0.5).add_(self._recv_buf, alpha=0.5)
+        g.unpack()
+        self.optimizer.step()
+
+    @torch.no_grad()
     def _step(self):
         if _core.size() == 1:
             self.optimizer.step()
+            return
+        if self.exchange == "rccl":
+            self._rccl_exchange()
             return
         if not self._init_done:
             # step 0: publish the initial model, then a barrier so every

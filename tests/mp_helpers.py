@@ -317,3 +317,25 @@ def dist_gloo_body(rank, np):
     out = float(t[0])
     kf.finalize()
     return out
+
+
+def pair_rccl_body(rank, np):
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.optimizers import PairAveragingOptimizer
+
+    kf.init()  # KUNGFU_TORCH_BACKEND=gloo via extra
+    torch.manual_seed(900 + rank)
+    lin = torch.nn.Linear(16, 4)
+    opt = PairAveragingOptimizer(
+        torch.optim.SGD(lin.parameters(), lr=0.0), exchange="rccl")
+    w0 = [p.detach().clone() for p in lin.parameters()]
+    for _ in range(3):
+        opt.zero_grad()
+        lin(torch.randn(4, 16)).sum().backward()
+        opt.step()
+    w1 = [round(float(p.flatten()[0]), 6) for p in lin.parameters()]
+    changed = any(not torch.equal(a, b.detach())
+                  for a, b in zip(w0, lin.parameters()))
+    kf.finalize()
+    return {"w": w1, "changed": changed}

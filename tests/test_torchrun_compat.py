@@ -100,3 +100,37 @@ def test_process_group_derived_store(port_block):
     extra = {"KUNGFU_TORCH_BACKEND": "gloo"}
     res = spawn_cluster(dist_gloo_body, 2, port_block, extra=extra)
     assert res == [3.0, 3.0]
+
+
+def test_pair_averaging_rccl_exchange(port_block):
+    """Symmetric tournament gossip over torch.distributed sendrecv
+    (gloo on CPU; RCCL on GPU boxes). With lr=0, after meeting every
+    other rank the replicas must mix toward the same average."""
+    from mp_helpers import pair_rccl_body, spawn_cluster
+
+    res = spawn_cluster(pair_rccl_body, 2, port_block,
+                        extra={"KUNGFU_TORCH_BACKEND": "gloo"})
+    assert all(r["changed"] for r in res)
+    # n=2: one round of pairwise averaging makes replicas identical
+    assert res[0]["w"] == res[1]["w"]
+
+
+def test_tournament_partner_schedule():
+    from kungfu_amd.optimizers.async_sgd import tournament_partner
+
+    for n in (2, 3, 4, 5, 8):
+        for s in range(2 * n):
+            seen = {}
+            for r in range(n):
+                p = tournament_partner(r, s, n)
+                seen[r] = p
+            for r, p in seen.items():
+                if p >= 0:
+                    assert seen[p] == r, (n, s, r, p, seen)  # symmetric
+            idle = [r for r, p in seen.items() if p < 0]
+            assert len(idle) == (n % 2)
+        # over n-1 rounds (even n) every rank meets every other rank
+        if n % 2 == 0:
+            for r in range(n):
+                met = {tournament_partner(r, s, n) for s in range(n - 1)}
+                assert met == set(range(n)) - {r}, (n, r, met)
