@@ -32,7 +32,7 @@ def parse_args():
                    choices=["resnet50", "vgg16", "inception-v3", "bert",
                             "slp"])
     p.add_argument("--optimizer", default="sync",
-                   choices=["sync", "sma", "pair", "gns"])
+                   choices=["sync", "sma", "pair", "gns", "sma-gns"])
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--dtype", default="bf16",
                    choices=["bf16", "bf16-pure", "fp32"],
@@ -162,6 +162,21 @@ def wrap_optimizer(args, model):
             fused_step=bool(args.fused_opt) and _t.cuda.is_available())
     if args.optimizer == "sma":
         return kfo.SynchronousAveragingOptimizer(inner)
+    if args.optimizer == "sma-gns":
+        # BASELINE config 4: SMA training + gradient-noise-scale monitor
+        opt = kfo.SynchronousAveragingOptimizer(inner)
+        probe = kfo.GradNoiseScaleProbe(
+            [p for g in inner.param_groups for p in g["params"]],
+            device_batch_size=args.batch_size, interval=5)
+        orig_step = opt._step
+
+        def stepped():
+            probe.observe()
+            orig_step()
+
+        opt._step = stepped
+        opt.gns_probe = probe
+        return opt
     if args.optimizer == "pair":
         import torch as _t
 

@@ -4,7 +4,7 @@ from kungfu_amd.optimizers.core import KungFuOptimizer  # noqa
 from kungfu_amd.optimizers.grad_variance import (  # noqa
     MonitorGradientVarianceOptimizer)
 from kungfu_amd.optimizers.noise_scale import (  # noqa
-    MonitorGradientNoiseScaleOptimizer)
+    GradNoiseScaleProbe, MonitorGradientNoiseScaleOptimizer)
 from kungfu_amd.optimizers.sma_sgd import (  # noqa
     SynchronousAveragingOptimizer)
 from kungfu_amd.optimizers.sync_sgd import SynchronousSGDOptimizer  # noqa

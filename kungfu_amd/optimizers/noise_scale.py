@@ -21,6 +21,70 @@ from kungfu_amd.optimizers.core import KungFuOptimizer
 from kungfu_amd.parallel.fusion import GradBucketReducer
 
 
+class GradNoiseScaleProbe:
+    """Standalone gradient-noise-scale monitor usable with ANY optimizer
+    (BASELINE config 4 pairs it with SynchronousAveraging, where gradients
+    are never all-reduced by the training path): every `interval` steps it
+    all-reduces a copy of the gradients to obtain the big-batch gradient
+    and updates the EMA estimators."""
+
+    def __init__(self, params, device_batch_size, alpha=0.6, interval=10):
+        import kungfu_amd  # noqa: F401 (ensure init)
+
+        self.params = [p for p in params if p.requires_grad]
+        self.b_small = float(device_batch_size)
+        self.alpha = float(alpha)
+        self.interval = max(1, int(interval))
+        self._steps = 0
+        self._ema_g2 = None
+        self._ema_s = None
+        self.noise_scale = float("nan")
+        self._probe = None
+
+    def _sqnorm(self, tensors):
+        if tensors[0].is_cuda:
+            from kungfu_amd.ops import hip as hip_ops
+
+            return float(hip_ops.norm2_multi(tensors).item())
+        return sum(float(t.float().pow(2).sum().item()) for t in tensors)
+
+    def observe(self):
+        """Call after backward() (before the optimizer consumes grads)."""
+        from kungfu_amd import _core
+        from kungfu_amd.ops import all_reduce
+
+        self._steps += 1
+        if _core.size() <= 1 or (self._steps - 1) % self.interval:
+            return self.noise_scale
+        grads = [p.grad for p in self.params if p.grad is not None]
+        if not grads:
+            return self.noise_scale
+        if self._probe is None:
+            total = sum(g.numel() for g in grads)
+            self._probe = torch.empty(total, dtype=grads[0].dtype,
+                                      device=grads[0].device)
+        off = 0
+        for g in grads:
+            self._probe[off:off + g.numel()].copy_(g.reshape(-1))
+            off += g.numel()
+        probe = self._probe[:off]
+        g_small_sq = self._sqnorm([probe])
+        all_reduce(probe, name="|gnsprobe", average=True)
+        g_big_sq = self._sqnorm([probe])
+        n = _core.size()
+        b_small, b_big = self.b_small, self.b_small * n
+        g2 = (b_big * g_big_sq - b_small * g_small_sq) / (b_big - b_small)
+        s = (g_small_sq - g_big_sq) / (1.0 / b_small - 1.0 / b_big)
+        a = self.alpha
+        self._ema_g2 = g2 if self._ema_g2 is None else (
+            a * g2 + (1 - a) * self._ema_g2)
+        self._ema_s = s if self._ema_s is None else (
+            a * s + (1 - a) * self._ema_s)
+        if self._ema_g2:
+            self.noise_scale = self._ema_s / self._ema_g2
+        return self.noise_scale
+
+
 class MonitorGradientNoiseScaleOptimizer(KungFuOptimizer):
     def __init__(self, optimizer, device_batch_size, alpha=0.6,
                  monitor_interval=1):

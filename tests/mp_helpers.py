@@ -61,6 +61,7 @@ def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None,
                 i, status, payload = q.get(timeout=timeout)
             except queue_mod.Empty:
                 import signal
+                import sys as sys_mod
                 import time as time_mod
 
                 for p in procs:  # dump stuck children's stacks to stderr
@@ -72,6 +73,16 @@ def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None,
                         except OSError:
                             pass
                 time_mod.sleep(2)
+                if _retry:
+                    # one retry on a fresh port block: protects against
+                    # stray listeners / load-induced slowness (a real
+                    # deadlock fails both attempts)
+                    print("[mp_helpers] cluster timed out; retrying on a "
+                          "fresh port block", file=sys_mod.stderr)
+                    for p in procs:
+                        p.terminate()
+                    return spawn_cluster(fn, np, base + 8192, strategy,
+                                         timeout, extra, _retry=False)
                 raise AssertionError("cluster timed out; results so far: %r"
                                      % (results,))
             if (status == "err" and _retry

@@ -178,3 +178,32 @@ def test_gradient_noise_scale(port_block):
 
 def test_gradient_variance(port_block):
     assert all(spawn_cluster(gvar_body, 2, port_block))
+
+
+def sma_gns_body(rank, np):
+    import math
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.optimizers import (GradNoiseScaleProbe,
+                                       SynchronousAveragingOptimizer)
+
+    kf.init(with_torch=False)
+    model = _make_model(seed=700)
+    opt = SynchronousAveragingOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.05), alpha=0.5)
+    probe = GradNoiseScaleProbe(model.parameters(), device_batch_size=8,
+                                interval=1)
+    torch.manual_seed(57 + rank)
+    for _ in range(3):
+        x = torch.randn(8, 32)
+        y = torch.randint(0, 4, (8,))
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        probe.observe()
+        opt.step()
+    kf.finalize()
+    return not math.isnan(probe.noise_scale)
+
+
+def test_sma_with_gns_probe(port_block):
+    assert all(spawn_cluster(sma_gns_body, 2, port_block))
