@@ -5,8 +5,6 @@ Self-contained transformer encoder with BERT-base dimensions (L=12, H=768,
 A=12, ~110M params); attention uses torch's fused SDPA, which lowers to the
 ROCm flash-attention path on MI355X.
 """
-import math
-
 import torch
 import torch.nn as nn
 import torch.nn.functional as F

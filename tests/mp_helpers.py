@@ -350,3 +350,36 @@ def pair_rccl_body(rank, np):
                   for a, b in zip(w0, lin.parameters()))
     kf.finalize()
     return {"w": w1, "changed": changed}
+
+
+def ops_wrappers_body(rank, np):
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd import ops
+
+    kf.init(with_torch=False)
+    out = {}
+    # torch-level wrappers on CPU tensors (routed through the C++ engine)
+    t = torch.full((5,), float(rank + 1))
+    ops.all_reduce(t, name="w1", average=True)
+    out["avg"] = float(t[0])
+    b = torch.full((3,), float(rank))
+    ops.broadcast(b, root=1, name="w2")
+    out["bcast_root1"] = float(b[0])
+    g = ops.all_gather(torch.tensor([float(rank)]), name="w3")
+    out["gather"] = g.flatten().tolist()
+    r = torch.full((2,), 1.0)
+    ops.reduce(r, name="w4")
+    out["reduced"] = float(r[0]) if rank == 0 else None
+    h = torch.full((4,), 1.0)
+    ops.hierarchical_all_reduce(h, name="w5")
+    out["hier"] = float(h[0])
+    # P2P tensor wrappers
+    ops.save_tensor("blob", torch.full((7,), float(rank + 10)))
+    kf.barrier()
+    dst = torch.zeros(7)
+    ok = ops.request_tensor((rank + 1) % np, "blob", dst)
+    out["p2p"] = ok and float(dst[0]) == float((rank + 1) % np + 10)
+    kf.barrier()
+    kf.finalize()
+    return out

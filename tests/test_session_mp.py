@@ -53,3 +53,16 @@ def test_monitoring(port_block):
         assert r["lat_len"] == 2
         assert r["egress_nonzero"]
         assert r["interference"] is False
+
+
+def test_torch_op_wrappers(port_block):
+    from mp_helpers import ops_wrappers_body
+
+    results = spawn_cluster(ops_wrappers_body, 2, port_block)
+    for r in results:
+        assert r["avg"] == pytest.approx(1.5)
+        assert r["bcast_root1"] == 1.0  # broadcast from root=1
+        assert r["gather"] == [0.0, 1.0]
+        assert r["hier"] == pytest.approx(2.0)
+        assert r["p2p"] is True
+    assert results[0]["reduced"] == pytest.approx(2.0)
