@@ -148,7 +148,7 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
                               unsigned short *__restrict__ y,
                               const float *__restrict__ a,
                               const float *__restrict__ b, long long M,
-                              int C)
+                              int C, unsigned char *__restrict__ mask)
 {
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
@@ -169,14 +169,21 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
         ushort8 rv;
         if (RES) rv = *(const ushort8 *)(res + base);
         ushort8 out;
+        unsigned char mb = 0;
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
             float u = fmaf(ar[k], b2f(v[k]), br[k]);
             if (RES) u += b2f(rv[k]);
-            if (RELU) u = fmaxf(u, 0.f);
+            if (RELU) {
+                if (u > 0.f) mb |= (unsigned char)(1u << k);
+                u = fmaxf(u, 0.f);
+            }
             out[k] = f2b(u);
         }
         *(ushort8 *)(y + base) = out;
+        // ReLU sign bits: 1 bit per element, so backward never re-reads
+        // the residual or recomputes the pre-activation
+        if (RELU && mask) mask[r * gpr + g] = mb;
     }
 }
 
@@ -184,12 +191,11 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
 // dy_m = dy * relu_mask (mask recomputed from x [+res]); xhat from
 // save_mean/save_rstd. out: f32[2*C] {sum_dy, sum_dyxhat} (pre-zeroed:
 // these ARE db and dw).
-template <bool RELU, bool RES>
+template <bool MASKED>
 __global__ void bn_bwd_reduce_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
-    const unsigned short *__restrict__ res,
-    const float *__restrict__ a, const float *__restrict__ b,
+    const unsigned char *__restrict__ mask,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     long long M, int C, float *__restrict__ sums)
 {
@@ -204,12 +210,10 @@ __global__ void bn_bwd_reduce_kernel(
     float s1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     float s2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     if (row_off < rows_per_blk) {
-        float ar[8], br[8], mr[8], rr[8];
+        float mr[8], rr[8];
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
             const int c = g * 8 + k;
-            ar[k] = a[c];
-            br[k] = b[c];
             mr[k] = mean[c];
             rr[k] = rstd[c];
         }
@@ -219,17 +223,13 @@ __global__ void bn_bwd_reduce_kernel(
             const long long base = r * C + (long long)g * 8;
             const ushort8 dv = *(const ushort8 *)(dy + base);
             const ushort8 xv = *(const ushort8 *)(x + base);
-            ushort8 rv;
-            if (RES) rv = *(const ushort8 *)(res + base);
+            const unsigned char mb =
+                MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
 #pragma unroll
             for (int k = 0; k < 8; ++k) {
                 const float xf = b2f(xv[k]);
                 float d = b2f(dv[k]);
-                if (RELU) {
-                    float u = fmaf(ar[k], xf, br[k]);
-                    if (RES) u += b2f(rv[k]);
-                    d = u > 0.f ? d : 0.f;
-                }
+                if (MASKED && !((mb >> k) & 1)) d = 0.f;
                 const float xh = (xf - mr[k]) * rr[k];
                 s1[k] += d;
                 s2[k] += d * xh;
@@ -260,12 +260,11 @@ __global__ void bn_fold_kernel(float *__restrict__ sums, int C)
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
 // dx = w*rstd * (dy_m - sum_dy/M - xhat * sum_dyxhat/M); d_res = dy_m.
-template <bool RELU, bool RES>
+template <bool MASKED, bool RES>
 __global__ void bn_bwd_dx_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
-    const unsigned short *__restrict__ res,
-    const float *__restrict__ a, const float *__restrict__ b,
+    const unsigned char *__restrict__ mask, const float *__restrict__ a,
     const float *__restrict__ mean, const float *__restrict__ rstd,
     const float *__restrict__ sums, long long M, int C,
     unsigned short *__restrict__ dx, unsigned short *__restrict__ dres)
@@ -276,12 +275,11 @@ __global__ void bn_bwd_dx_kernel(
     const int row_off = threadIdx.x / gpr;
     if (row_off >= rows_per_blk) return;
     const float invM = 1.f / (float)M;
-    float ar[8], br[8], mr[8], rr[8], t1[8], t2[8];
+    float ar[8], mr[8], rr[8], t1[8], t2[8];
 #pragma unroll
     for (int k = 0; k < 8; ++k) {
         const int c = g * 8 + k;
         ar[k] = a[c];
-        br[k] = b[c];
         mr[k] = mean[c];
         rr[k] = rstd[c];
         t1[k] = sums[c] * invM;      // mean of dy_m
@@ -293,18 +291,14 @@ __global__ void bn_bwd_dx_kernel(
         const long long base = r * C + (long long)g * 8;
         const ushort8 dv = *(const ushort8 *)(dy + base);
         const ushort8 xv = *(const ushort8 *)(x + base);
-        ushort8 rv;
-        if (RES) rv = *(const ushort8 *)(res + base);
+        const unsigned char mb =
+            MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
         ushort8 dxo, dro;
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
             const float xf = b2f(xv[k]);
             float d = b2f(dv[k]);
-            if (RELU) {
-                float u = fmaf(ar[k], xf, br[k]);
-                if (RES) u += b2f(rv[k]);
-                d = u > 0.f ? d : 0.f;
-            }
+            if (MASKED && !((mb >> k) & 1)) d = 0.f;
             if (RES) dro[k] = f2b(d);
             const float xh = (xf - mr[k]) * rr[k];
             const float t = d - t1[k] - xh * t2[k];
@@ -358,7 +352,7 @@ hipError_t kf_bn_finalize(const void *sums, const void *weight,
 
 hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
                      const void *b, long long M, int C, int relu,
-                     void *stream)
+                     void *mask, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int rows_per_blk = BLOCK / (C / 8);
@@ -371,7 +365,8 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
     hipLaunchKernelGGL((bn_fwd_kernel<R, E>), grid, block, 0, s,            \
                        (const unsigned short *)x,                           \
                        (const unsigned short *)res, (unsigned short *)y,    \
-                       (const float *)a, (const float *)b, M, C)
+                       (const float *)a, (const float *)b, M, C,            \
+                       (unsigned char *)mask)
     if (relu && res) CASE(true, true);
     else if (relu) CASE(true, false);
     else if (res) CASE(false, true);
@@ -380,9 +375,9 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
     return hipGetLastError();
 }
 
-hipError_t kf_bn_bwd_reduce(const void *dy, const void *x, const void *res,
-                            const void *a, const void *b, const void *mean,
-                            const void *rstd, long long M, int C, int relu,
+hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
+                            const void *mask, const void *mean,
+                            const void *rstd, long long M, int C,
                             void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
@@ -396,17 +391,14 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x, const void *res,
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
     const size_t lds = 2 * C * sizeof(float);
-#define CASE(R, E)                                                          \
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<R, E>), grid, block, lds, s,   \
+#define CASE(MK)                                                            \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<MK>), grid, block, lds, s,     \
                        (const unsigned short *)dy,                          \
                        (const unsigned short *)x,                           \
-                       (const unsigned short *)res, (const float *)a,       \
-                       (const float *)b, (const float *)mean,               \
+                       (const unsigned char *)mask, (const float *)mean,    \
                        (const float *)rstd, M, C, (float *)sums)
-    if (relu && res) CASE(true, true);
-    else if (relu) CASE(true, false);
-    else if (res) CASE(false, true);
-    else CASE(false, false);
+    if (mask) CASE(true);
+    else CASE(false);
 #undef CASE
     return hipGetLastError();
 }
@@ -418,10 +410,10 @@ hipError_t kf_bn_fold(void *sums, int C, void *stream)
     return hipGetLastError();
 }
 
-hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *res,
-                        const void *a, const void *b, const void *mean,
-                        const void *rstd, const void *sums, long long M,
-                        int C, int relu, void *dx, void *dres, void *stream)
+hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *mask,
+                        const void *a, const void *mean, const void *rstd,
+                        const void *sums, long long M, int C, void *dx,
+                        void *dres, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int rows_per_blk = BLOCK / (C / 8);
@@ -430,17 +422,17 @@ hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *res,
     if (blocks < 1) blocks = 1;
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
-#define CASE(R, E)                                                          \
-    hipLaunchKernelGGL((bn_bwd_dx_kernel<R, E>), grid, block, 0, s,         \
+#define CASE(MK, E)                                                         \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<MK, E>), grid, block, 0, s,        \
                        (const unsigned short *)dy,                          \
                        (const unsigned short *)x,                           \
-                       (const unsigned short *)res, (const float *)a,       \
-                       (const float *)b, (const float *)mean,               \
-                       (const float *)rstd, (const float *)sums, M, C,      \
-                       (unsigned short *)dx, (unsigned short *)dres)
-    if (relu && res) CASE(true, true);
-    else if (relu) CASE(true, false);
-    else if (res) CASE(false, true);
+                       (const unsigned char *)mask, (const float *)a,       \
+                       (const float *)mean, (const float *)rstd,            \
+                       (const float *)sums, M, C, (unsigned short *)dx,     \
+                       (unsigned short *)dres)
+    if (mask && dres) CASE(true, true);
+    else if (mask) CASE(true, false);
+    else if (dres) CASE(false, true);
     else CASE(false, false);
 #undef CASE
     return hipGetLastError();

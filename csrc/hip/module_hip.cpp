@@ -22,16 +22,15 @@ hipError_t kf_bn_finalize(const void *, const void *, const void *, void *,
                           void *, void *, void *, void *, void *, long long,
                           int, float, float, void *);
 hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
-                     const void *, long long, int, int, void *);
+                     const void *, long long, int, int, void *, void *);
 hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
-                            const void *, const void *, const void *,
-                            const void *, long long, int, int, void *,
-                            void *);
+                            const void *, const void *, long long, int,
+                            void *, void *);
 hipError_t kf_bn_fold(void *, int, void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
-                        const void *, const void *, long long, int, int,
-                        void *, void *, void *);
+                        const void *, long long, int, void *, void *,
+                        void *);
 hipError_t kf_unpack(const void *, int, const void *, float, int, void *);
 hipError_t kf_avg_inplace(void *, const void *, float, long long, int,
                           void *);
@@ -215,38 +214,37 @@ PYBIND11_MODULE(_hip, m)
           });
     m.def("bn_fwd",
           [](uintptr_t x, uintptr_t res, uintptr_t y, uintptr_t a,
-             uintptr_t b, long long M, int C, bool relu, uintptr_t stream) {
+             uintptr_t b, long long M, int C, bool relu, uintptr_t mask,
+             uintptr_t stream) {
               check(kf_bn_fwd((const void *)x, (const void *)res,
                               (void *)y, (const void *)a, (const void *)b,
-                              M, C, relu ? 1 : 0, (void *)stream),
+                              M, C, relu ? 1 : 0, (void *)mask,
+                              (void *)stream),
                     "kf_bn_fwd");
           });
     m.def("bn_bwd_reduce",
-          [](uintptr_t dy, uintptr_t x, uintptr_t res, uintptr_t a,
-             uintptr_t b, uintptr_t mean, uintptr_t rstd, long long M,
-             int C, bool relu, uintptr_t sums, uintptr_t stream) {
+          [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t mean,
+             uintptr_t rstd, long long M, int C, uintptr_t sums,
+             uintptr_t stream) {
               check(kf_bn_bwd_reduce((const void *)dy, (const void *)x,
-                                     (const void *)res, (const void *)a,
-                                     (const void *)b, (const void *)mean,
+                                     (const void *)mask,
+                                     (const void *)mean,
                                      (const void *)rstd, M, C,
-                                     relu ? 1 : 0, (void *)sums,
-                                     (void *)stream),
+                                     (void *)sums, (void *)stream),
                     "kf_bn_bwd_reduce");
           });
     m.def("bn_fold", [](uintptr_t sums, int C, uintptr_t stream) {
         check(kf_bn_fold((void *)sums, C, (void *)stream), "kf_bn_fold");
     });
     m.def("bn_bwd_dx",
-          [](uintptr_t dy, uintptr_t x, uintptr_t res, uintptr_t a,
-             uintptr_t b, uintptr_t mean, uintptr_t rstd, uintptr_t sums,
-             long long M, int C, bool relu, uintptr_t dx, uintptr_t dres,
-             uintptr_t stream) {
+          [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t a,
+             uintptr_t mean, uintptr_t rstd, uintptr_t sums, long long M,
+             int C, uintptr_t dx, uintptr_t dres, uintptr_t stream) {
               check(kf_bn_bwd_dx((const void *)dy, (const void *)x,
-                                 (const void *)res, (const void *)a,
-                                 (const void *)b, (const void *)mean,
-                                 (const void *)rstd, (const void *)sums, M,
-                                 C, relu ? 1 : 0, (void *)dx, (void *)dres,
-                                 (void *)stream),
+                                 (const void *)mask, (const void *)a,
+                                 (const void *)mean, (const void *)rstd,
+                                 (const void *)sums, M, C, (void *)dx,
+                                 (void *)dres, (void *)stream),
                     "kf_bn_bwd_dx");
           });
 }

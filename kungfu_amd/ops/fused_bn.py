@@ -55,37 +55,41 @@ class _FusedBNFunction(torch.autograd.Function):
             b = (bias - running_mean * a).float()
             save_mean, save_rstd = running_mean, rstd
         y = torch.empty_like(x)
+        mask = None
+        if relu and training:
+            # 1 bit/elem ReLU sign mask: backward never re-reads the
+            # residual or recomputes the pre-activation
+            mask = torch.empty(M * (C // 8), dtype=torch.uint8, device=dev)
         _hip.bn_fwd(x.data_ptr(),
                     residual.data_ptr() if residual is not None else 0,
                     y.data_ptr(), a.data_ptr(), b.data_ptr(), M, C,
-                    bool(relu), s)
-        ctx.save_for_backward(x, residual if residual is not None else x,
-                              a, b, save_mean, save_rstd, weight)
+                    bool(relu), mask.data_ptr() if mask is not None else 0,
+                    s)
+        ctx.save_for_backward(x, a, save_mean, save_rstd,
+                              mask if mask is not None else x)
         ctx.has_res = residual is not None
-        ctx.relu = bool(relu)
+        ctx.has_mask = mask is not None
         ctx.MC = (M, C)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, residual, a, b, save_mean, save_rstd, weight = ctx.saved_tensors
+        x, a, save_mean, save_rstd, mask = ctx.saved_tensors
         M, C = ctx.MC
         s = _stream()
         dy = dy.contiguous(memory_format=torch.channels_last)
         dev = x.device
         sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
-        res_ptr = residual.data_ptr() if ctx.has_res else 0
-        _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), res_ptr,
-                           a.data_ptr(), b.data_ptr(),
+        mask_ptr = mask.data_ptr() if ctx.has_mask else 0
+        _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), mask_ptr,
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
-                           C, ctx.relu, sums.data_ptr(), s)
+                           C, sums.data_ptr(), s)
         _hip.bn_fold(sums.data_ptr(), C, s)  # fold shadow accumulators
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
-        _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), res_ptr, a.data_ptr(),
-                       b.data_ptr(), save_mean.data_ptr(),
-                       save_rstd.data_ptr(), sums.data_ptr(), M, C,
-                       ctx.relu, dx.data_ptr(),
+        _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), mask_ptr, a.data_ptr(),
+                       save_mean.data_ptr(), save_rstd.data_ptr(),
+                       sums.data_ptr(), M, C, dx.data_ptr(),
                        dres.data_ptr() if dres is not None else 0, s)
         db = sums[:C]               # db = sum(dy_m)
         dw = sums[C:2 * C]          # dw = sum(dy_m * xhat)
