@@ -274,7 +274,7 @@ def main():
     elapsed = time.perf_counter() - t0
     elapsed = max_over_ranks(elapsed)
 
-    n_gpus = world if use_cuda else world  # one rank per GPU
+    n_gpus = world  # one rank per GPU
     total_items = args.steps * per_gpu_batch * world
     value = total_items / elapsed
     unit = "images/sec" if args.model != "bert" else "sequences/sec"

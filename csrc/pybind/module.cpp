@@ -48,15 +48,10 @@ struct Tracer {
         static Tracer t;
         static bool init = [] {
             const char *e = std::getenv("KUNGFU_ENABLE_TRACE");
-            get_unsafe().enabled = e && *e && std::string(e) != "0";
+            t.enabled = e && *e && std::string(e) != "0";
             return true;
         }();
         (void)init;
-        return get_unsafe();
-    }
-    static Tracer &get_unsafe()
-    {
-        static Tracer t;
         return t;
     }
     void record(const char *what, double start_us, double dur_us)

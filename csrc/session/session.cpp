@@ -272,13 +272,12 @@ bool Session::consensus(const void *data, size_t len,
 {
     // agree on length first, then min/max over the bytes
     uint64_t lmin = len, lmax = len;
+    uint64_t local_len = len;
     Workspace wl;
     wl.count = 1;
     wl.dt = DType::U64;
     wl.name = name + "|len";
-    wl.send = &len;
-    uint64_t tmp = len;
-    wl.send = &tmp;
+    wl.send = &local_len;
     wl.recv = &lmin;
     wl.op = ReduceOp::MIN;
     run_strategies(wl, global_, false);
