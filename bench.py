@@ -112,7 +112,9 @@ def build_model_and_data(args, device, dtype, amp):
     if args.model == "bert":
         from kungfu_amd.models import bert_base
 
-        model = bert_base(max_len=max(args.seq_len, 128)).to(
+        use_fused_ln = amp and device.type == "cuda"
+        model = bert_base(max_len=max(args.seq_len, 128),
+                          fused_ln=use_fused_ln).to(
             device=device, dtype=dtype)
         if amp:
             model = model.to(dtype=torch.float32)

@@ -1,0 +1,236 @@
+// Fused LayerNorm for bf16 activations with fp32 parameters — gfx950.
+//
+// Why: under torch autocast, LayerNorm runs in fp32, inserting
+// bf16<->f32 cast passes around all 24 LNs of BERT-base (measured
+// 1.6 ms of a 12.7 ms step) and a multi-kernel backward. These kernels
+// keep activations bf16 end to end and accumulate in f32.
+//
+// Shape: x is [N rows][H] with H % 8 == 0 (hidden size, e.g. 768/3072).
+// One wave (64 lanes) owns one row: vec-8 loads land each row in
+// registers once; mean/var via __shfl_down tree; the normalize reuses the
+// registers (single global read in forward). Backward is single-pass:
+// per-row dx from wave-reduced stats, per-column dW/db partials
+// accumulated in LDS and flushed through NSHADOW interleaved global
+// accumulators (same contention pattern as fused_bn.hip).
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace {
+
+constexpr int WAVE = 64;
+constexpr int BLOCK = 256;              // 4 waves = 4 rows per block
+constexpr int ROWS_PER_BLOCK = BLOCK / WAVE;
+constexpr int NSHADOW = 8;
+constexpr int MAX_K = 8;  // supports H <= 8 * 512 = 4096
+
+typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+
+__device__ inline float b2f(unsigned short u)
+{
+    union {
+        unsigned int i;
+        float f;
+    } v;
+    v.i = (unsigned int)u << 16;
+    return v.f;
+}
+
+__device__ inline unsigned short f2b(float f)
+{
+    union {
+        unsigned int i;
+        float f;
+    } v;
+    v.f = f;
+    unsigned int r = v.i + 0x7fff + ((v.i >> 16) & 1);
+    return (unsigned short)(r >> 16);
+}
+
+__device__ inline float wave_sum(float v)
+{
+    for (int off = WAVE / 2; off > 0; off >>= 1) v += __shfl_down(v, off);
+    return __shfl(v, 0);
+}
+
+// ---- forward: y = (x - mean) * rstd * w + b ----
+__global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
+                              unsigned short *__restrict__ y,
+                              const float *__restrict__ w,
+                              const float *__restrict__ b,
+                              float *__restrict__ save_mean,
+                              float *__restrict__ save_rstd, long long N,
+                              int H, float eps)
+{
+    const int lane = threadIdx.x % WAVE;
+    const int wid = threadIdx.x / WAVE;
+    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);  // vec8 iters per lane
+    const long long row_step = (long long)gridDim.x * ROWS_PER_BLOCK;
+    for (long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid; r < N;
+         r += row_step) {
+        const unsigned short *xr = x + r * H;
+        ushort8 v[MAX_K];
+        float s = 0.f, q = 0.f;
+        for (int k = 0; k < K; ++k) {
+            const int h0 = (k * WAVE + lane) * 8;
+            if (h0 < H) {
+                v[k] = *(const ushort8 *)(xr + h0);
+#pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const float f = b2f(v[k][i]);
+                    s += f;
+                    q += f * f;
+                }
+            }
+        }
+        s = wave_sum(s);
+        q = wave_sum(q);
+        const float mean = s / (float)H;
+        const float var = fmaxf(q / (float)H - mean * mean, 0.f);
+        const float rstd = rsqrtf(var + eps);
+        if (lane == 0) {
+            save_mean[r] = mean;
+            save_rstd[r] = rstd;
+        }
+        for (int k = 0; k < K; ++k) {
+            const int h0 = (k * WAVE + lane) * 8;
+            if (h0 < H) {
+                ushort8 o;
+#pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const float xh = (b2f(v[k][i]) - mean) * rstd;
+                    o[i] = f2b(fmaf(xh, w[h0 + i], b[h0 + i]));
+                }
+                *(ushort8 *)(y + r * H + h0) = o;
+            }
+        }
+    }
+}
+
+// ---- backward (single pass) ----
+// dx = rstd * (g - mean(g) - xhat * mean(g * xhat)),  g = dy * w
+// dW[h] += sum_rows dy * xhat ; db[h] += sum_rows dy  (via LDS + shadows)
+__global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
+                              const unsigned short *__restrict__ x,
+                              const float *__restrict__ w,
+                              const float *__restrict__ save_mean,
+                              const float *__restrict__ save_rstd,
+                              long long N, int H,
+                              unsigned short *__restrict__ dx,
+                              float *__restrict__ wb_sums)
+{
+    extern __shared__ float lds[];  // 2*H floats {dw, db}
+    const int lane = threadIdx.x % WAVE;
+    const int wid = threadIdx.x / WAVE;
+    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
+    for (int i = threadIdx.x; i < 2 * H; i += BLOCK) lds[i] = 0.f;
+    __syncthreads();
+
+    const long long row_step = (long long)gridDim.x * ROWS_PER_BLOCK;
+    for (long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid; r < N;
+         r += row_step) {
+        const unsigned short *dyr = dy + r * H;
+        const unsigned short *xr = x + r * H;
+        const float mean = save_mean[r];
+        const float rstd = save_rstd[r];
+        ushort8 dv[MAX_K], xv[MAX_K];
+        float s1 = 0.f, s2 = 0.f;
+        for (int k = 0; k < K; ++k) {
+            const int h0 = (k * WAVE + lane) * 8;
+            if (h0 < H) {
+                dv[k] = *(const ushort8 *)(dyr + h0);
+                xv[k] = *(const ushort8 *)(xr + h0);
+#pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const float d = b2f(dv[k][i]);
+                    const float xh = (b2f(xv[k][i]) - mean) * rstd;
+                    const float g = d * w[h0 + i];
+                    s1 += g;
+                    s2 += g * xh;
+                }
+            }
+        }
+        s1 = wave_sum(s1) / (float)H;
+        s2 = wave_sum(s2) / (float)H;
+        for (int k = 0; k < K; ++k) {
+            const int h0 = (k * WAVE + lane) * 8;
+            if (h0 < H) {
+                ushort8 o;
+#pragma unroll
+                for (int i = 0; i < 8; ++i) {
+                    const float d = b2f(dv[k][i]);
+                    const float xh = (b2f(xv[k][i]) - mean) * rstd;
+                    const float g = d * w[h0 + i];
+                    o[i] = f2b(rstd * (g - s1 - xh * s2));
+                    atomicAdd(&lds[h0 + i], d * xh);      // dW partial
+                    atomicAdd(&lds[H + h0 + i], d);       // db partial
+                }
+                *(ushort8 *)(dx + r * H + h0) = o;
+            }
+        }
+    }
+    __syncthreads();
+    float *shadow = wb_sums + (size_t)(blockIdx.x % NSHADOW) * 2 * H;
+    for (int i = threadIdx.x; i < 2 * H; i += BLOCK) {
+        atomicAdd(&shadow[i], lds[i]);
+    }
+}
+
+// Fold the shadow copies of {dw, db} into copy 0.
+__global__ void ln_fold_kernel(float *__restrict__ wb_sums, int H)
+{
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= 2 * H) return;
+    float acc = wb_sums[i];
+    for (int k = 1; k < NSHADOW; ++k) acc += wb_sums[(size_t)k * 2 * H + i];
+    wb_sums[i] = acc;
+}
+
+}  // namespace
+
+extern "C" {
+
+hipError_t kf_ln_fwd(const void *x, void *y, const void *w, const void *b,
+                     void *save_mean, void *save_rstd, long long N, int H,
+                     float eps, void *stream)
+{
+    if (H % 8 != 0 || H > WAVE * 8 * MAX_K) return hipErrorInvalidValue;
+    long long blocks = (N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(ln_fwd_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
+                       0, (hipStream_t)stream, (const unsigned short *)x,
+                       (unsigned short *)y, (const float *)w,
+                       (const float *)b, (float *)save_mean,
+                       (float *)save_rstd, N, H, eps);
+    return hipGetLastError();
+}
+
+hipError_t kf_ln_bwd(const void *dy, const void *x, const void *w,
+                     const void *save_mean, const void *save_rstd,
+                     long long N, int H, void *dx, void *wb_sums,
+                     void *stream)
+{
+    if (H % 8 != 0 || H > WAVE * 8 * MAX_K) return hipErrorInvalidValue;
+    long long blocks = (N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
+    const long long by_bytes = (N * H * 4 + 65535) / 65536;
+    if (blocks > by_bytes) blocks = by_bytes;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    hipLaunchKernelGGL(ln_bwd_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
+                       2 * H * sizeof(float), (hipStream_t)stream,
+                       (const unsigned short *)dy, (const unsigned short *)x,
+                       (const float *)w, (const float *)save_mean,
+                       (const float *)save_rstd, N, H, (unsigned short *)dx,
+                       (float *)wb_sums);
+    return hipGetLastError();
+}
+
+hipError_t kf_ln_fold(void *wb_sums, int H, void *stream)
+{
+    hipLaunchKernelGGL(ln_fold_kernel, dim3((2 * H + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (float *)wb_sums, H);
+    return hipGetLastError();
+}
+
+}  // extern "C"

@@ -41,6 +41,11 @@ hipError_t kf_dot(const void *, const void *, long long, void *, int,
 hipError_t kf_sgd_momentum(void *, const void *, void *, long long, float,
                            float, float, float, int, int, void *);
 hipError_t kf_transform2(void *, const void *, long long, int, int, void *);
+hipError_t kf_ln_fwd(const void *, void *, const void *, const void *,
+                     void *, void *, long long, int, float, void *);
+hipError_t kf_ln_bwd(const void *, const void *, const void *, const void *,
+                     const void *, long long, int, void *, void *, void *);
+hipError_t kf_ln_fold(void *, int, void *);
 }
 
 namespace {
@@ -191,6 +196,30 @@ PYBIND11_MODULE(_hip, m)
                     "kf_transform2");
           });
     m.def("device_synchronize", [] { check(hipDeviceSynchronize(), "sync"); });
+
+    // ---- fused LayerNorm (bf16 activations, fp32 params) ----
+    m.def("ln_fwd",
+          [](uintptr_t x, uintptr_t y, uintptr_t w, uintptr_t b,
+             uintptr_t smean, uintptr_t srstd, long long N, int H,
+             float eps, uintptr_t stream) {
+              check(kf_ln_fwd((const void *)x, (void *)y, (const void *)w,
+                              (const void *)b, (void *)smean,
+                              (void *)srstd, N, H, eps, (void *)stream),
+                    "kf_ln_fwd");
+          });
+    m.def("ln_bwd",
+          [](uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t smean,
+             uintptr_t srstd, long long N, int H, uintptr_t dx,
+             uintptr_t wb_sums, uintptr_t stream) {
+              check(kf_ln_bwd((const void *)dy, (const void *)x,
+                              (const void *)w, (const void *)smean,
+                              (const void *)srstd, N, H, (void *)dx,
+                              (void *)wb_sums, (void *)stream),
+                    "kf_ln_bwd");
+          });
+    m.def("ln_fold", [](uintptr_t wb, int H, uintptr_t stream) {
+        check(kf_ln_fold((void *)wb, H, (void *)stream), "kf_ln_fold");
+    });
 
     // ---- fused BatchNorm(+residual+ReLU), NHWC bf16 ----
     m.def("bn_stats",

@@ -10,16 +10,24 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 
+def _make_ln(hidden, fused):
+    if fused:
+        from kungfu_amd.ops.fused_ln import FusedLayerNorm
+
+        return FusedLayerNorm(hidden)
+    return nn.LayerNorm(hidden)
+
+
 class BertLayer(nn.Module):
-    def __init__(self, hidden, heads, ffn):
+    def __init__(self, hidden, heads, ffn, fused_ln=False):
         super().__init__()
         self.heads = heads
         self.qkv = nn.Linear(hidden, 3 * hidden)
         self.proj = nn.Linear(hidden, hidden)
-        self.ln1 = nn.LayerNorm(hidden)
+        self.ln1 = _make_ln(hidden, fused_ln)
         self.fc1 = nn.Linear(hidden, ffn)
         self.fc2 = nn.Linear(ffn, hidden)
-        self.ln2 = nn.LayerNorm(hidden)
+        self.ln2 = _make_ln(hidden, fused_ln)
 
     def forward(self, x):
         b, s, h = x.shape
@@ -34,13 +42,15 @@ class BertLayer(nn.Module):
 
 class Bert(nn.Module):
     def __init__(self, vocab=30522, hidden=768, layers=12, heads=12,
-                 ffn=3072, max_len=512):
+                 ffn=3072, max_len=512, fused_ln=False):
         super().__init__()
+        self.fused_ln = fused_ln
         self.tok = nn.Embedding(vocab, hidden)
         self.pos = nn.Embedding(max_len, hidden)
-        self.ln = nn.LayerNorm(hidden)
+        self.ln = _make_ln(hidden, fused_ln)
         self.blocks = nn.ModuleList(
-            [BertLayer(hidden, heads, ffn) for _ in range(layers)])
+            [BertLayer(hidden, heads, ffn, fused_ln=fused_ln)
+             for _ in range(layers)])
         self.head = nn.Linear(hidden, vocab)
         for m in self.modules():
             if isinstance(m, nn.Linear):
@@ -52,11 +62,14 @@ class Bert(nn.Module):
     def forward(self, ids):
         b, s = ids.shape
         pos = torch.arange(s, device=ids.device).unsqueeze(0)
-        x = self.ln(self.tok(ids) + self.pos(pos))
+        x = self.tok(ids) + self.pos(pos)
+        if self.fused_ln and x.is_cuda:
+            x = x.to(torch.bfloat16)  # embeddings stay fp32 under autocast
+        x = self.ln(x)
         for blk in self.blocks:
             x = blk(x)
         return self.head(x)
 
 
-def bert_base(vocab=30522, max_len=512):
-    return Bert(vocab=vocab, max_len=max_len)
+def bert_base(vocab=30522, max_len=512, fused_ln=False):
+    return Bert(vocab=vocab, max_len=max_len, fused_ln=fused_ln)

@@ -35,6 +35,7 @@ CORE_SOURCES = [
 HIP_SOURCES = [
     "csrc/hip/kernels.hip",
     "csrc/hip/fused_bn.hip",
+    "csrc/hip/fused_ln.hip",
     "csrc/hip/module_hip.cpp",
 ]
 

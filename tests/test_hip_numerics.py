@@ -195,3 +195,52 @@ def test_fused_resnet_step_matches_eager_loss():
     l1.backward()
     l2.backward()
     torch.cuda.synchronize()
+
+
+@pytest.mark.parametrize("shape", [(32, 768), (4, 128, 768), (7, 3072)])
+def test_fused_layernorm_vs_eager(shape):
+    from kungfu_amd.ops.fused_ln import FusedLayerNorm
+
+    torch.manual_seed(2)
+    H = shape[-1]
+    x1 = (torch.randn(*shape, device="cuda") * 1.5 + 0.2).to(
+        torch.bfloat16).requires_grad_()
+    x2 = x1.detach().clone().requires_grad_()
+    m = FusedLayerNorm(H).to("cuda")
+    m.weight.data.uniform_(0.5, 1.5)
+    m.bias.data.uniform_(-0.3, 0.3)
+
+    y1 = m(x1)
+    y2 = torch.nn.functional.layer_norm(x2.float(), (H,), m.weight,
+                                        m.bias, m.eps)
+    torch.cuda.synchronize()
+    assert torch.allclose(y1.float(), y2, atol=3e-2, rtol=3e-2)
+
+    g = torch.randn_like(y2)
+    y1.backward(g.to(torch.bfloat16))
+    gw1, gb1 = m.weight.grad.clone(), m.bias.grad.clone()
+    m.weight.grad = None
+    m.bias.grad = None
+    y2.backward(g)
+    torch.cuda.synchronize()
+    assert torch.allclose(x1.grad.float(), x2.grad.float(), atol=5e-2,
+                          rtol=5e-2)
+    assert torch.allclose(gw1, m.weight.grad, rtol=2e-2, atol=2e-1)
+    assert torch.allclose(gb1, m.bias.grad, rtol=2e-2, atol=2e-1)
+
+
+def test_bert_fused_ln_step():
+    import kungfu_amd as kf
+    from kungfu_amd.models import bert_base
+
+    kf.init()
+    m = bert_base(max_len=128, fused_ln=True).to("cuda")
+    ids = torch.randint(0, 30522, (2, 64), device="cuda")
+    labels = torch.randint(0, 30522, (2, 64), device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        out = m(ids)
+        loss = torch.nn.functional.cross_entropy(out.flatten(0, 1),
+                                                 labels.flatten())
+    loss.backward()
+    torch.cuda.synchronize()
+    assert float(loss) == float(loss)
