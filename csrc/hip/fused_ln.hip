@@ -54,6 +54,9 @@ __device__ inline float wave_sum(float v)
 }
 
 // ---- forward: y = (x - mean) * rstd * w + b ----
+// K (vec8 iterations per lane) is a template parameter: runtime-indexed
+// register arrays would be allocated in scratch memory (200x slower).
+template <int K>
 __global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
                               unsigned short *__restrict__ y,
                               const float *__restrict__ w,
@@ -64,13 +67,13 @@ __global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
 {
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
-    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);  // vec8 iters per lane
     const long long row_step = (long long)gridDim.x * ROWS_PER_BLOCK;
     for (long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid; r < N;
          r += row_step) {
         const unsigned short *xr = x + r * H;
-        ushort8 v[MAX_K];
+        ushort8 v[K];
         float s = 0.f, q = 0.f;
+#pragma unroll
         for (int k = 0; k < K; ++k) {
             const int h0 = (k * WAVE + lane) * 8;
             if (h0 < H) {
@@ -92,6 +95,7 @@ __global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
             save_mean[r] = mean;
             save_rstd[r] = rstd;
         }
+#pragma unroll
         for (int k = 0; k < K; ++k) {
             const int h0 = (k * WAVE + lane) * 8;
             if (h0 < H) {
@@ -110,6 +114,7 @@ __global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
 // ---- backward (single pass) ----
 // dx = rstd * (g - mean(g) - xhat * mean(g * xhat)),  g = dy * w
 // dW[h] += sum_rows dy * xhat ; db[h] += sum_rows dy  (via LDS + shadows)
+template <int K>
 __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
                               const unsigned short *__restrict__ x,
                               const float *__restrict__ w,
@@ -122,10 +127,14 @@ __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
     extern __shared__ float lds[];  // 2*H floats {dw, db}
     const int lane = threadIdx.x % WAVE;
     const int wid = threadIdx.x / WAVE;
-    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
     for (int i = threadIdx.x; i < 2 * H; i += BLOCK) lds[i] = 0.f;
     __syncthreads();
 
+    // per-lane register accumulators for the column sums: every row
+    // visits the SAME h-positions for a given (lane, k), so dW/db
+    // partials stay in registers and hit LDS only once per block
+    float accw[K][8] = {};
+    float accb[K][8] = {};
     const long long row_step = (long long)gridDim.x * ROWS_PER_BLOCK;
     for (long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid; r < N;
          r += row_step) {
@@ -133,8 +142,9 @@ __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
         const unsigned short *xr = x + r * H;
         const float mean = save_mean[r];
         const float rstd = save_rstd[r];
-        ushort8 dv[MAX_K], xv[MAX_K];
+        ushort8 dv[K], xv[K];
         float s1 = 0.f, s2 = 0.f;
+#pragma unroll
         for (int k = 0; k < K; ++k) {
             const int h0 = (k * WAVE + lane) * 8;
             if (h0 < H) {
@@ -152,6 +162,7 @@ __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
         }
         s1 = wave_sum(s1) / (float)H;
         s2 = wave_sum(s2) / (float)H;
+#pragma unroll
         for (int k = 0; k < K; ++k) {
             const int h0 = (k * WAVE + lane) * 8;
             if (h0 < H) {
@@ -162,10 +173,21 @@ __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
                     const float xh = (b2f(xv[k][i]) - mean) * rstd;
                     const float g = d * w[h0 + i];
                     o[i] = f2b(rstd * (g - s1 - xh * s2));
-                    atomicAdd(&lds[h0 + i], d * xh);      // dW partial
-                    atomicAdd(&lds[H + h0 + i], d);       // db partial
+                    accw[k][i] += d * xh;
+                    accb[k][i] += d;
                 }
                 *(ushort8 *)(dx + r * H + h0) = o;
+            }
+        }
+    }
+#pragma unroll
+    for (int k = 0; k < K; ++k) {
+        const int h0 = (k * WAVE + lane) * 8;
+        if (h0 < H) {
+#pragma unroll
+            for (int i = 0; i < 8; ++i) {
+                atomicAdd(&lds[h0 + i], accw[k][i]);
+                atomicAdd(&lds[H + h0 + i], accb[k][i]);
             }
         }
     }
@@ -198,11 +220,23 @@ hipError_t kf_ln_fwd(const void *x, void *y, const void *w, const void *b,
     long long blocks = (N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
     if (blocks > 4096) blocks = 4096;
     if (blocks < 1) blocks = 1;
-    hipLaunchKernelGGL(ln_fwd_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
-                       0, (hipStream_t)stream, (const unsigned short *)x,
-                       (unsigned short *)y, (const float *)w,
-                       (const float *)b, (float *)save_mean,
-                       (float *)save_rstd, N, H, eps);
+    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
+#define CASE(KK)                                                            \
+    hipLaunchKernelGGL((ln_fwd_kernel<KK>), dim3((uint32_t)blocks),         \
+                       dim3(BLOCK), 0, (hipStream_t)stream,                 \
+                       (const unsigned short *)x, (unsigned short *)y,      \
+                       (const float *)w, (const float *)b,                  \
+                       (float *)save_mean, (float *)save_rstd, N, H, eps)
+    switch (K) {
+    case 1: CASE(1); break;
+    case 2: CASE(2); break;
+    case 3: CASE(3); break;
+    case 4: CASE(4); break;
+    case 6: CASE(6); break;
+    case 8: CASE(8); break;
+    default: return hipErrorInvalidValue;
+    }
+#undef CASE
     return hipGetLastError();
 }
 
@@ -217,12 +251,24 @@ hipError_t kf_ln_bwd(const void *dy, const void *x, const void *w,
     if (blocks > by_bytes) blocks = by_bytes;
     if (blocks > 2048) blocks = 2048;
     if (blocks < 1) blocks = 1;
-    hipLaunchKernelGGL(ln_bwd_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
-                       2 * H * sizeof(float), (hipStream_t)stream,
-                       (const unsigned short *)dy, (const unsigned short *)x,
-                       (const float *)w, (const float *)save_mean,
-                       (const float *)save_rstd, N, H, (unsigned short *)dx,
-                       (float *)wb_sums);
+    const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
+#define CASE(KK)                                                            \
+    hipLaunchKernelGGL((ln_bwd_kernel<KK>), dim3((uint32_t)blocks),         \
+                       dim3(BLOCK), 2 * H * sizeof(float),                  \
+                       (hipStream_t)stream, (const unsigned short *)dy,     \
+                       (const unsigned short *)x, (const float *)w,         \
+                       (const float *)save_mean, (const float *)save_rstd,  \
+                       N, H, (unsigned short *)dx, (float *)wb_sums)
+    switch (K) {
+    case 1: CASE(1); break;
+    case 2: CASE(2); break;
+    case 3: CASE(3); break;
+    case 4: CASE(4); break;
+    case 6: CASE(6); break;
+    case 8: CASE(8); break;
+    default: return hipErrorInvalidValue;
+    }
+#undef CASE
     return hipGetLastError();
 }
 
