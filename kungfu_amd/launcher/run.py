@@ -35,6 +35,9 @@ def parse_args(argv=None):
     p.add_argument("-port", type=int, default=38080, help="runner port")
     p.add_argument("-port-range", dest="port_base", type=int, default=31100,
                    help="first worker port")
+    p.add_argument("-P", dest="peers", default=None,
+                   help="explicit worker peer list ip:port,... "
+                        "(overrides -H/-np port allocation)")
     p.add_argument("-strategy", default="AUTO",
                    help="STAR|MULTI_STAR|RING|CLIQUE|TREE|BINARY_TREE|"
                         "BINARY_TREE_STAR|MULTI_BINARY_TREE_STAR|AUTO")
@@ -54,6 +57,12 @@ def parse_args(argv=None):
     p.add_argument("-logdir", default=None)
     p.add_argument("-logfile", default=None)
     p.add_argument("-delay", default=None)
+    p.add_argument("-t0", default=None,
+                   help="job start timestamp override")
+    p.add_argument("-u", dest="ssh_user", default=None,
+                   help="ssh user (kungfu-distribute)")
+    p.add_argument("-debug-port", type=int, default=0,
+                   help="accepted for reference-CLI compatibility")
     p.add_argument("-allow-nvlink", action="store_true",
                    help="accepted for reference-CLI compatibility (no-op: "
                         "xGMI is always used on MI355X)")
@@ -116,7 +125,11 @@ class Runner:
         from kungfu_amd import _core
 
         self.core = _core
-        self.peers = _core.gen_peer_list(hosts, args.np, args.port_base)
+        if args.peers:
+            self.peers = args.peers
+        else:
+            self.peers = _core.gen_peer_list(hosts, args.np,
+                                             args.port_base)
         self.runners = _core.gen_runner_list(hosts, args.port)
         self.version = args.init_version
         self.procs = {}  # spec -> Proc
@@ -143,7 +156,8 @@ class Runner:
         env["KUNGFU_ALLREDUCE_STRATEGY"] = self.args.strategy
         if self.config_server_url():
             env["KUNGFU_CONFIG_SERVER"] = self.config_server_url()
-        env["KUNGFU_JOB_START_TIMESTAMP"] = str(int(time.time()))
+        env["KUNGFU_JOB_START_TIMESTAMP"] = (
+            self.args.t0 or str(int(time.time())))
         env["KUNGFU_MONITOR_PORT"] = str(self.args.monitor_port)
         # GPU slot assignment: local rank among this host's workers
         ip = spec.rsplit(":", 1)[0]
