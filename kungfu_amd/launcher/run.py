@@ -111,6 +111,7 @@ class Proc:
 class Runner:
     def __init__(self, args):
         self.args = args
+        self._logf = None
         if args.nic:
             args.self_ip = infer_self_ip(args.nic)
         if args.hostfile:
@@ -186,6 +187,8 @@ class Runner:
             logf = open(os.path.join(
                 logdir, "%s@%d.log" % (spec.replace(":", "."), version)),
                 "ab")
+        elif self.args.logfile:
+            logf = self._shared_logfile()
 
         def stream(src, dst):
             prefix = ("\x1b[%dm[%s]\x1b[0m " % (color, spec)).encode()
@@ -206,6 +209,11 @@ class Runner:
         with self.lock:
             self.procs[spec] = proc
         return proc
+
+    def _shared_logfile(self):
+        if self._logf is None:
+            self._logf = open(self.args.logfile, "ab")
+        return self._logf
 
     def kill(self, spec):
         with self.lock:
