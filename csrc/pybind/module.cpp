@@ -11,6 +11,8 @@
 #include <chrono>
 #include <cstring>
 #include <deque>
+#include <future>
+#include <map>
 #include <memory>
 #include <mutex>
 #include <tuple>
@@ -26,6 +28,37 @@ namespace {
 
 std::unique_ptr<Peer> g_peer;
 std::mutex g_mu;
+
+// Async collective handles (reference: HandleManager int-handle wait map,
+// include/kungfu/utils/handler_manager.hpp + all_reduce_cuda_async).
+// Each async op runs on its own thread; wait() joins and rethrows.
+struct AsyncOps {
+    std::mutex mu;
+    uint64_t next = 1;
+    std::map<uint64_t, std::future<void>> ops;
+
+    uint64_t launch(std::function<void()> fn)
+    {
+        std::lock_guard<std::mutex> lk(mu);
+        const uint64_t h = next++;
+        ops[h] = std::async(std::launch::async, std::move(fn));
+        return h;
+    }
+    void wait(uint64_t h)
+    {
+        std::future<void> f;
+        {
+            std::lock_guard<std::mutex> lk(mu);
+            auto it = ops.find(h);
+            if (it == ops.end())
+                throw std::runtime_error("unknown async handle");
+            f = std::move(it->second);
+            ops.erase(it);
+        }
+        f.get();  // rethrows the op's exception, if any
+    }
+};
+AsyncOps g_async;
 
 Peer &peer()
 {
@@ -277,6 +310,35 @@ PYBIND11_MODULE(_core, m)
               py::gil_scoped_release rel;
               peer().session().cross_all_reduce(w);
           });
+    // ---- async variants (returns a handle; wait(handle) joins) ----
+    m.def("all_reduce_async",
+          [](uintptr_t sp, uintptr_t r, size_t count, int dt, int op,
+             const std::string &name) {
+              auto w = make_ws(sp, r, count, dt, op, name);
+              return g_async.launch([w] {
+                  StallGuard sg("all_reduce_async");
+                  peer().session().all_reduce(w);
+              });
+          });
+    m.def("broadcast_async",
+          [](uintptr_t sp, uintptr_t r, size_t count, int dt,
+             const std::string &name, int root) {
+              auto w = make_ws(sp, r, count, dt, 0, name);
+              return g_async.launch(
+                  [w, root] { peer().session().broadcast(w, root); });
+          });
+    m.def("all_gather_async",
+          [](uintptr_t sp, uintptr_t r, size_t count, int dt,
+             const std::string &name) {
+              auto w = make_ws(sp, r, count, dt, 0, name);
+              return g_async.launch(
+                  [w] { peer().session().all_gather(w); });
+          });
+    m.def("wait_handle", [](uint64_t h) {
+        py::gil_scoped_release rel;
+        g_async.wait(h);
+    });
+
     m.def("consensus", [](py::bytes data, const std::string &name) {
         std::string s = data;
         py::gil_scoped_release rel;

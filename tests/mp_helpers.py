@@ -383,3 +383,30 @@ def ops_wrappers_body(rank, np):
     kf.barrier()
     kf.finalize()
     return out
+
+
+def async_ops_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    # several async all-reduces in flight (different names), then wait all
+    bufs = [np_.full(10_000, float(rank + 1 + i), dtype=np_.float32)
+            for i in range(4)]
+    handles = [
+        _core.all_reduce_async(b.ctypes.data, b.ctypes.data, b.size, 10, 0,
+                               "as%d" % i) for i, b in enumerate(bufs)
+    ]
+    for h in handles:
+        _core.wait_handle(h)
+    expect = [sum(r + 1 + i for r in range(np)) for i in range(4)]
+    ok = all(float(b[0]) == e for b, e in zip(bufs, expect))
+    # async broadcast + gather
+    g = np_.array([float(rank)], dtype=np_.float32)
+    go = np_.zeros(np, dtype=np_.float32)
+    h = _core.all_gather_async(g.ctypes.data, go.ctypes.data, 1, 10, "ag")
+    _core.wait_handle(h)
+    ok = ok and go.tolist() == [float(i) for i in range(np)]
+    kf.finalize()
+    return ok

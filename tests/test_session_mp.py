@@ -66,3 +66,9 @@ def test_torch_op_wrappers(port_block):
         assert r["hier"] == pytest.approx(2.0)
         assert r["p2p"] is True
     assert results[0]["reduced"] == pytest.approx(2.0)
+
+
+def test_async_handles(port_block):
+    from mp_helpers import async_ops_body
+
+    assert spawn_cluster(async_ops_body, 3, port_block) == [True] * 3
