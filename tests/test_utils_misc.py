@@ -116,3 +116,25 @@ def test_prim_mst():
     assert parent[1] == 0
     assert parent[2] == 1
     assert parent[3] == 0
+
+
+def test_elastic_sampler_progress():
+    """set_progress continues the shard from a synced offset (reference
+    datasets/adaptor.py skip+shard)."""
+    import kungfu_amd as kf
+    from kungfu_amd.data import ElasticShardSampler
+
+    kf.init(with_torch=False)  # single process: rank 0 of 1
+    s = ElasticShardSampler(20, seed=3)
+    full = list(s)
+    assert len(full) == 20
+    s.set_progress(12)
+    rest = list(s)
+    assert rest == full[12:]
+    assert len(s) == 8
+
+
+def test_cli_tools_exist():
+    for t in ("kungfu-run", "kungfu-config-server", "kungfu-distribute"):
+        p = os.path.join(ROOT, "tools", t)
+        assert os.path.exists(p) and os.access(p, os.X_OK)
