@@ -410,3 +410,27 @@ def async_ops_body(rank, np):
     ok = ok and go.tolist() == [float(i) for i in range(np)]
     kf.finalize()
     return ok
+
+
+def gpu_pair_store_body(rank, np):
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.models import SLP
+    from kungfu_amd.optimizers import PairAveragingOptimizer
+
+    kf.init(with_torch=False)  # store-mode gossip: no process group needed
+    torch.manual_seed(800 + rank)
+    m = SLP(in_features=64, classes=8).to("cuda")
+    opt = PairAveragingOptimizer(
+        torch.optim.SGD(m.parameters(), lr=0.01),
+        peer_selection="roundrobin")
+    for _ in range(3):
+        x = torch.randn(8, 1, 8, 8, device="cuda")
+        y = torch.randint(0, 8, (8,), device="cuda")
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(m(x), y).backward()
+        opt.step()
+    torch.cuda.synchronize()
+    kf.barrier()
+    kf.finalize()
+    return True
