@@ -79,8 +79,21 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
     float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
     if (row_off < rows_per_blk) {
         const long long row_step = (long long)gridDim.x * rows_per_blk;
-        for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-             r < M; r += row_step) {
+        long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+        // 2-row ILP: two independent 16-B loads in flight per iteration
+        for (; r + row_step < M; r += 2 * row_step) {
+            const ushort8 va = *(const ushort8 *)(x + r * C + g * 8);
+            const ushort8 vb =
+                *(const ushort8 *)(x + (r + row_step) * C + g * 8);
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                const float fa = b2f(va[k]);
+                const float fb = b2f(vb[k]);
+                s[k] += fa + fb;
+                q[k] += fa * fa + fb * fb;
+            }
+        }
+        for (; r < M; r += row_step) {
             const ushort8 v = *(const ushort8 *)(x + r * C + g * 8);
 #pragma unroll
             for (int k = 0; k < 8; ++k) {
@@ -218,8 +231,33 @@ __global__ void bn_bwd_reduce_kernel(
             rr[k] = rstd[c];
         }
         const long long row_step = (long long)gridDim.x * rows_per_blk;
-        for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-             r < M; r += row_step) {
+        long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+        // 2-row ILP: four independent 16-B loads in flight per iteration
+        for (; r + row_step < M; r += 2 * row_step) {
+            const long long ba = r * C + (long long)g * 8;
+            const long long bb = (r + row_step) * C + (long long)g * 8;
+            const ushort8 dva = *(const ushort8 *)(dy + ba);
+            const ushort8 xva = *(const ushort8 *)(x + ba);
+            const ushort8 dvb = *(const ushort8 *)(dy + bb);
+            const ushort8 xvb = *(const ushort8 *)(x + bb);
+            const unsigned char ma =
+                MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
+            const unsigned char mbm =
+                MASKED ? mask[(r + row_step) * gpr + g]
+                       : (unsigned char)0xff;
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                float da = b2f(dva[k]);
+                float db = b2f(dvb[k]);
+                if (MASKED && !((ma >> k) & 1)) da = 0.f;
+                if (MASKED && !((mbm >> k) & 1)) db = 0.f;
+                const float xha = (b2f(xva[k]) - mr[k]) * rr[k];
+                const float xhb = (b2f(xvb[k]) - mr[k]) * rr[k];
+                s1[k] += da + db;
+                s2[k] += da * xha + db * xhb;
+            }
+        }
+        for (; r < M; r += row_step) {
             const long long base = r * C + (long long)g * 8;
             const ushort8 dv = *(const ushort8 *)(dy + base);
             const ushort8 xv = *(const ushort8 *)(x + base);

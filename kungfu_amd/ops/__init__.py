@@ -221,6 +221,27 @@ def request_tensor(target_rank, name, out):
     return ok
 
 
+_egress_last = {}
+
+
+def egress_rates(window_hint=None):
+    """Per-peer egress rates in bytes/sec since the previous call
+    (reference: EgressRates op, ops/cpu/monitoring.cpp + monitor rates)."""
+    import time
+
+    global _egress_last
+    _ensure_init()
+    now = time.time()
+    cur = _core.egress_bytes()
+    rates = {}
+    prev, t0 = _egress_last.get("snap", ({}, now))
+    dt = max(now - t0, 1e-6)
+    for peer_str, total in cur.items():
+        rates[peer_str] = (total - prev.get(peer_str, 0)) / dt
+    _egress_last["snap"] = (cur, now)
+    return rates
+
+
 # ---- adaptive topology ----
 
 def get_peer_latencies():

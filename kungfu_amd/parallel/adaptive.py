@@ -63,3 +63,17 @@ def strategy_throughputs():
     """Per-strategy monitored throughput stats (bytes/sec)."""
     _ensure_init()
     return _core.strategy_stats()
+
+
+def print_strategy_stats():
+    """Log per-strategy monitored throughput (reference PrintStategyStats,
+    session/adaptiveStrategies.go)."""
+    from kungfu_amd import rank
+
+    stats = strategy_throughputs()
+    for i, s in enumerate(stats):
+        print("[kungfu rank %d] strategy %d: ops=%d bytes=%d "
+              "throughput=%.1f MB/s" %
+              (rank(), i, s["ops"], s["bytes"], s["throughput"] / 1e6),
+              flush=True)
+    return stats
