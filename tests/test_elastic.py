@@ -66,3 +66,19 @@ def test_elastic_grow_and_shrink(port_block):
     assert len(done) == 2
     assert all("size=2 step=8" in ln for ln in done)
     assert out.count("DETACHED") == 2
+
+
+def test_elastic_churn(port_block):
+    """Repeated grow/shrink churn: 2 -> 4 -> 2 -> 3 -> 2 in one run."""
+    r = run_watch([
+        "-np", "2", "-w", "-port", str(port_block), "-port-range",
+        str(port_block + 2), "-builtin-config-port", str(port_block + 1),
+        sys.executable, "examples/elastic_trainer.py",
+        "--schedule", "2:4,4:2,6:3,8:2", "--max-step", "10",
+    ], timeout=300)
+    out = _plain(r.stdout)
+    assert r.returncode == 0, out + _plain(r.stderr)
+    done = [ln for ln in out.splitlines() if "DONE" in ln]
+    assert len(done) == 2, out
+    assert all("size=2 step=10" in ln for ln in done)
+    assert out.count("RESIZED") >= 6  # at least surviving workers log each
