@@ -244,3 +244,26 @@ def test_bert_fused_ln_step():
     loss.backward()
     torch.cuda.synchronize()
     assert float(loss) == float(loss)
+
+
+def test_fused_bn_eval_mode():
+    """Inference path: normalize with running stats, no stat updates."""
+    from kungfu_amd.ops.fused_bn import FusedBNReLU2d
+
+    torch.manual_seed(4)
+    m = FusedBNReLU2d(64, relu=True).to("cuda")
+    m.running_mean.uniform_(-0.2, 0.2)
+    m.running_var.uniform_(0.8, 1.2)
+    m.weight.data.uniform_(0.5, 1.5)
+    m.bias.data.uniform_(-0.2, 0.2)
+    m.eval()
+    x = (torch.randn(2, 64, 8, 8, device="cuda")).to(
+        torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    rm, rv = m.running_mean.clone(), m.running_var.clone()
+    with torch.no_grad():
+        y = m(x)
+        ref = torch.relu(torch.nn.functional.batch_norm(
+            x.float(), rm, rv, m.weight, m.bias, False, 0.1, m.eps))
+    torch.cuda.synchronize()
+    assert torch.allclose(y.float(), ref, atol=5e-2, rtol=5e-2)
+    assert torch.equal(m.running_mean, rm)  # eval must not update stats
