@@ -434,3 +434,38 @@ def gpu_pair_store_body(rank, np):
     kf.barrier()
     kf.finalize()
     return True
+
+
+def dtype_sweep_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    # (numpy dtype, core code, op code, expected for 2 ranks r=0,1)
+    cases = [
+        ("uint8", 0, 0, 3),        # (1+r) summed
+        ("int8", 1, 3, 2),         # (1+r) product
+        ("int16", 2, 0, 3),
+        ("int32", 3, 1, 1),        # min of 1+r
+        ("int64", 4, 2, 2),        # max
+        ("uint16", 5, 0, 3),
+        ("uint32", 6, 0, 3),
+        ("uint64", 7, 0, 3),
+        ("float32", 10, 0, 3.0),
+        ("float64", 11, 3, 2.0),
+    ]
+    ok = True
+    for name, code, op, expect in cases:
+        a = np_.full(13, rank + 1, dtype=np_.dtype(name))
+        _core.all_reduce(a.ctypes.data, a.ctypes.data, a.size, code, op,
+                         "dt-" + name)
+        ok = ok and a[0] == expect
+    # f16 via torch
+    import torch
+
+    th = torch.full((9,), float(rank + 1), dtype=torch.float16)
+    _core.all_reduce(th.data_ptr(), th.data_ptr(), 9, 8, 0, "dt-f16")
+    ok = ok and float(th[0]) == 3.0
+    kf.finalize()
+    return ok

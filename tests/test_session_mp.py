@@ -72,3 +72,9 @@ def test_async_handles(port_block):
     from mp_helpers import async_ops_body
 
     assert spawn_cluster(async_ops_body, 3, port_block) == [True] * 3
+
+
+def test_all_dtypes_reduce(port_block):
+    from mp_helpers import dtype_sweep_body
+
+    assert spawn_cluster(dtype_sweep_body, 2, port_block) == [True, True]
