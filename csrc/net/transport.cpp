@@ -119,9 +119,12 @@ void Conn::close_fd()
 
 // ---------- Server ----------
 
-std::string Server::unix_sock_path(uint16_t port)
+std::string Server::unix_sock_path(const PeerID &peer)
 {
-    return "/tmp/kungfu-amd-" + std::to_string(port) + ".sock";
+    // keyed by ip AND port: loopback aliases (127.0.0.x multi-host
+    // simulation) may reuse port numbers on one machine
+    return "/tmp/kungfu-amd-" + std::to_string(peer.ipv4) + "-" +
+           std::to_string(peer.port) + ".sock";
 }
 
 Server::Server(const PeerID &self, bool use_unix)
@@ -143,12 +146,18 @@ void Server::start(FrameHandler handler, std::function<bool(uint32_t)> token_ok)
     ::setsockopt(tcp_fd_, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
     sockaddr_in addr{};
     addr.sin_family = AF_INET;
-    addr.sin_addr.s_addr = htonl(INADDR_ANY);
+    // bind the advertised IP so several logical hosts can coexist on one
+    // machine (e.g. 127.0.0.1 vs 127.0.0.2 loopback aliases); fall back
+    // to ANY when the address is not a local interface (NAT/public IPs)
+    addr.sin_addr.s_addr = htonl(self_.ipv4);
     addr.sin_port = htons(self_.port);
-    if (::bind(tcp_fd_, (sockaddr *)&addr, sizeof(addr)) != 0)
-        throw std::runtime_error("bind failed on port " +
-                                 std::to_string(self_.port) + ": " +
-                                 std::strerror(errno));
+    if (::bind(tcp_fd_, (sockaddr *)&addr, sizeof(addr)) != 0) {
+        addr.sin_addr.s_addr = htonl(INADDR_ANY);
+        if (::bind(tcp_fd_, (sockaddr *)&addr, sizeof(addr)) != 0)
+            throw std::runtime_error("bind failed on port " +
+                                     std::to_string(self_.port) + ": " +
+                                     std::strerror(errno));
+    }
     if (::listen(tcp_fd_, 128) != 0)
         throw std::runtime_error("listen failed");
     {
@@ -163,7 +172,7 @@ void Server::start(FrameHandler handler, std::function<bool(uint32_t)> token_ok)
         if (unix_fd_ >= 0) {
             sockaddr_un ua{};
             ua.sun_family = AF_UNIX;
-            std::string path = unix_sock_path(self_.port);
+            std::string path = unix_sock_path(self_);
             ::unlink(path.c_str());
             std::snprintf(ua.sun_path, sizeof(ua.sun_path), "%s",
                           path.c_str());
@@ -260,7 +269,7 @@ void Server::stop()
     if (unix_fd_ >= 0) {
         ::shutdown(unix_fd_, SHUT_RDWR);
         ::close(unix_fd_);
-        ::unlink(unix_sock_path(self_.port).c_str());
+        ::unlink(unix_sock_path(self_).c_str());
     }
     std::vector<std::thread> ts;
     {
@@ -298,7 +307,7 @@ std::shared_ptr<Conn> Client::get_conn(const PeerID &remote, ConnType type,
                 sockaddr_un ua{};
                 ua.sun_family = AF_UNIX;
                 std::snprintf(ua.sun_path, sizeof(ua.sun_path), "%s",
-                              Server::unix_sock_path(remote.port).c_str());
+                              Server::unix_sock_path(remote).c_str());
                 if (::connect(fd, (sockaddr *)&ua, sizeof(ua)) == 0) {
                     set_bufsizes(fd);
                     break;

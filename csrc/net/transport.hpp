@@ -80,7 +80,7 @@ class Server {
     void start(FrameHandler handler,
                std::function<bool(uint32_t)> token_ok);
     void stop();
-    static std::string unix_sock_path(uint16_t port);
+    static std::string unix_sock_path(const PeerID &peer);
 
   private:
     void accept_loop(int listen_fd);

@@ -258,8 +258,7 @@ class Runner:
     def local_specs(self, peers_csv):
         ip = self.args.self_ip
         return [s for s in peers_csv.split(",")
-                if s.rsplit(":", 1)[0] in (ip, "127.0.0.1")
-                or ip == "127.0.0.1"]
+                if s.rsplit(":", 1)[0] == ip]
 
     def simple_run(self):
         """Static mode (reference SimpleRun): spawn all local procs, wait."""

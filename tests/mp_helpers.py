@@ -469,3 +469,89 @@ def dtype_sweep_body(rank, np):
     ok = ok and float(th[0]) == 3.0
     kf.finalize()
     return ok
+
+
+def set_env_multihost(i, np, base, strategy="AUTO", hosts=2):
+    """Simulated multi-host cluster on loopback: rank i lives on
+    127.0.0.(1 + i % hosts) — the whole 127/8 block routes to lo, so
+    cross-'host' traffic takes the TCP (non-unix, non-shm) paths."""
+    peers = ",".join("127.0.0.%d:%d" % (1 + j % hosts, base + j)
+                     for j in range(np))
+    os.environ["KUNGFU_SELF_SPEC"] = "127.0.0.%d:%d" % (1 + i % hosts,
+                                                        base + i)
+    os.environ["KUNGFU_INIT_PEERS"] = peers
+    os.environ["KUNGFU_ALLREDUCE_STRATEGY"] = strategy
+    os.environ["KUNGFU_NO_UNIX_SOCK"] = "1"
+    for k in ("KUNGFU_CONFIG_SERVER", "MASTER_ADDR", "MASTER_PORT"):
+        os.environ.pop(k, None)
+
+
+def run_worker_mh(fn, i, np, base, strategy, q, hosts):
+    try:
+        import faulthandler
+        import signal
+
+        faulthandler.enable()
+        faulthandler.register(signal.SIGUSR1, all_threads=True)
+        set_env_multihost(i, np, base, strategy, hosts)
+        q.put((i, "ok", fn(i, np)))
+    except Exception as e:  # pragma: no cover
+        import traceback
+
+        q.put((i, "err", "%s\n%s" % (e, traceback.format_exc())))
+
+
+def spawn_multihost(fn, np, base, strategy="AUTO", hosts=2, timeout=120):
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=run_worker_mh,
+                         args=(fn, i, np, base, strategy, q, hosts))
+             for i in range(np)]
+    for p in procs:
+        p.start()
+    results = {}
+    try:
+        for _ in range(np):
+            i, status, payload = q.get(timeout=timeout)
+            assert status == "ok", "rank %d failed: %s" % (i, payload)
+            results[i] = payload
+    finally:
+        for p in procs:
+            p.join(timeout=15)
+            if p.is_alive():
+                p.terminate()
+    return [results[i] for i in range(np)]
+
+
+def multihost_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    out = {"hosts": _core.host_count(), "local_size": _core.local_size()}
+    # global all-reduce across 'hosts' (chunked, hierarchical topologies)
+    x = np_.arange(300_000, dtype=np_.float32) + rank
+    _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "mh")
+    expect0 = np * 0 + sum(range(np))
+    assert float(x[0]) == expect0, (float(x[0]), expect0)
+    # hierarchical: local reduce -> cross -> local bcast
+    h = np_.full(50_000, 1.0, dtype=np_.float32)
+    _core.local_reduce(h.ctypes.data, h.ctypes.data, h.size, 10, 0, "lr")
+    _core.cross_all_reduce(h.ctypes.data, h.ctypes.data, h.size, 10, 0,
+                           "cx")
+    _core.local_broadcast(h.ctypes.data, h.ctypes.data, h.size, 10, "lb")
+    out["hier"] = float(h[0])
+    # P2P across 'hosts': inline-payload branch (no shm for remote IPs)
+    blob = np_.full(5000, float(rank + 40), dtype=np_.float32)
+    _core.save("mhmodel", blob.ctypes.data, blob.nbytes)
+    kf.barrier()
+    got = np_.zeros_like(blob)
+    target = (rank + 1) % np  # neighbor is on the other 'host'
+    assert _core.request(target, "mhmodel", got.ctypes.data, got.nbytes)
+    out["p2p"] = float(got[0]) == float(target + 40)
+    kf.barrier()
+    kf.finalize()
+    return out

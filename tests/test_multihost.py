@@ -1,0 +1,29 @@
+"""Runtime multi-host simulation: ranks on 127.0.0.1 vs 127.0.0.2 (the
+whole 127/8 block is loopback), so host grouping, cross-host topologies,
+hierarchical local/cross collectives, and the non-colocated P2P inline
+path all execute for real (reference pattern: cluster-in-docker tests,
+.github/workflows/cluster.yaml)."""
+import pytest
+
+from mp_helpers import multihost_body, spawn_multihost
+
+
+@pytest.mark.parametrize("strategy", ["AUTO", "BINARY_TREE_STAR", "RING",
+                                      "MULTI_BINARY_TREE_STAR",
+                                      "MULTI_STAR", "TREE"])
+def test_two_hosts_collectives(strategy, port_block):
+    results = spawn_multihost(multihost_body, 4, port_block, strategy,
+                              hosts=2)
+    for r in results:
+        assert r["hosts"] == 2
+        assert r["local_size"] == 2
+        assert r["hier"] == pytest.approx(4.0)
+        assert r["p2p"] is True
+
+
+def test_three_hosts(port_block):
+    results = spawn_multihost(multihost_body, 6, port_block, "AUTO",
+                              hosts=3)
+    for r in results:
+        assert r["hosts"] == 3
+        assert r["hier"] == pytest.approx(6.0)
