@@ -69,7 +69,7 @@ class MonitorState:
             return min(self.epochs.values())
 
 
-def start_monitor_server(state, port=MONITOR_PORT):
+def start_monitor_server(state, port=MONITOR_PORT, host="0.0.0.0"):
     class Handler(BaseHTTPRequestHandler):
         def log_message(self, *a):
             pass
@@ -85,7 +85,7 @@ def start_monitor_server(state, port=MONITOR_PORT):
             self.send_header("Content-Length", "0")
             self.end_headers()
 
-    srv = ThreadingHTTPServer(("0.0.0.0", port), Handler)
+    srv = ThreadingHTTPServer((host, port), Handler)
     t = threading.Thread(target=srv.serve_forever, daemon=True)
     t.start()
     return srv
@@ -129,7 +129,10 @@ def monitored_run(runner, grace=GRACE_SECONDS):
     (reference monitored.go:18-75)."""
     grace = grace or GRACE_SECONDS
     state = MonitorState()
-    srv = start_monitor_server(state, runner.args.monitor_port)
+    # bind the runner's own IP so one monitor per simulated host can
+    # coexist on a single machine (loopback aliases)
+    srv = start_monitor_server(state, runner.args.monitor_port,
+                               runner.args.self_ip)
     world = len(runner.peers.split(","))
     try:
         attempt = 0
@@ -165,10 +168,15 @@ def monitored_run(runner, grace=GRACE_SECONDS):
                 time.sleep(0.5)
             if failed:
                 attempt += 1
+                locally_detected = state.other_down_epoch is None
                 min_epoch = state.min_epoch()
                 print("[kungfu-run] failure detected (attempt %d); "
-                      "restarting from epoch %d" % (attempt, min_epoch))
-                _broadcast_otherdown(runner, min_epoch)
+                      "restarting from epoch %d" % (attempt, min_epoch),
+                      flush=True)
+                if locally_detected:
+                    # only the detecting host fans out: echoing an
+                    # otherdown back would ping-pong restarts
+                    _broadcast_otherdown(runner, min_epoch)
                 with runner.lock:
                     specs = list(runner.procs.keys())
                 for s in specs:

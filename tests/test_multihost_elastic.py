@@ -64,3 +64,38 @@ def test_two_runners_elastic_grow(port_block):
         assert len(resized) == 4  # the four survivors each re-synced
     finally:
         srv.shutdown()
+
+
+def test_two_runners_auto_recover(port_block):
+    """Cross-host failure recovery: a worker on host 2 crashes; host 1's
+    monitor detects the stall (all heartbeats go to host 0 = runners[0]),
+    broadcasts otherdown to host 2's monitor, and BOTH runners restart
+    their workers with adjusted epochs (reference monitored.go +
+    monitor.go otherdown flow)."""
+    hosts = "127.0.0.1:2,127.0.0.2:2"
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    ckpt = "/tmp/kf_mh_fr_%d.pt" % port_block
+    procs = []
+    for self_ip in ("127.0.0.1", "127.0.0.2"):
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "kungfu_amd.run",
+             "-np", "4", "-H", hosts, "-self", self_ip,
+             "-port", str(port_block), "-port-range", str(port_block + 2),
+             "-auto-recover", "3s", "-monitor-port", str(port_block + 58),
+             sys.executable, "examples/failure_recovery_trainer.py",
+             "--n-epochs", "4", "--crash-at-epoch", "2", "--ckpt", ckpt],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=240)
+        outs.append((p.returncode, _plain(out), _plain(err)))
+    all_out = "\n".join(o for _, o, _ in outs)
+    for rc, out, err in outs:
+        assert rc == 0, all_out + err
+    assert "CRASHING rank 0 now" in all_out
+    # both runners restarted their workers
+    assert all_out.count("RESTARTED from epoch 2") == 4, all_out
+    ends = [ln for ln in all_out.splitlines() if "TRAIN END" in ln]
+    assert len(ends) == 4 and all("total_epochs=4" in ln for ln in ends)
