@@ -79,3 +79,22 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def replace_cluster(new_workers):
+    """Test helper: PUT an arbitrary worker list to the config server
+    (exercises resizes the schedule API cannot express, e.g. removing
+    rank 0)."""
+    import json
+    import os
+    import urllib.request
+
+    url = os.environ["KUNGFU_CONFIG_SERVER"]
+    if not url.startswith("http"):
+        url = "http://" + url
+    body = json.dumps({
+        "runners": os.environ.get("KUNGFU_INIT_RUNNERS", "").split(","),
+        "workers": new_workers,
+    }).encode()
+    req = urllib.request.Request(url + "/config", data=body, method="PUT")
+    urllib.request.urlopen(req, timeout=5)
