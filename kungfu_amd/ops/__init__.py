@@ -137,20 +137,82 @@ def cpu_staged_all_reduce(tensor, op="sum", name=None):
     return tensor
 
 
+_hier_groups = {}
+
+
+def _hierarchical_groups():
+    """Build (once per cluster version) the torch.distributed sub-groups
+    for hierarchical all-reduce: one group per host, plus one group of the
+    local masters. Every rank must enter every new_group call
+    (torch.distributed contract), so group construction is itself a
+    collective."""
+    import torch.distributed as dist
+
+    ver = int(_core.cluster_version())
+    cached = _hier_groups.get("v")
+    if cached is not None and cached[0] == ver:
+        return cached[1]
+    import json
+
+    # host layout from the control plane: allgather each rank's host ip
+    world = _core.size()
+    my_host = _core.uid().rsplit(":", 1)[0]
+    import numpy as np
+
+    enc = np.zeros(16, dtype=np.uint8)
+    raw = my_host.encode()[:16]
+    enc[:len(raw)] = np.frombuffer(raw, dtype=np.uint8)
+    out = np.zeros(16 * world, dtype=np.uint8)
+    _core.all_gather(enc.ctypes.data, out.ctypes.data, 16, 0, "|hiermap")
+    hosts = [bytes(out[i * 16:(i + 1) * 16]).rstrip(b"\0").decode()
+             for i in range(world)]
+    order = []
+    for h in hosts:
+        if h not in order:
+            order.append(h)
+    local_groups = {}
+    masters = []
+    for h in order:
+        ranks = [r for r in range(world) if hosts[r] == h]
+        masters.append(ranks[0])
+        local_groups[h] = dist.new_group(ranks=ranks)
+    cross_group = dist.new_group(ranks=masters)
+    info = {
+        "local": local_groups[my_host],
+        "local_master": min(r for r in range(world)
+                            if hosts[r] == my_host),
+        "is_master": _core.rank() in masters,
+        "cross": cross_group,
+        "json": json.dumps({"hosts": hosts, "masters": masters}),
+    }
+    _hier_groups["v"] = (ver, info)
+    return info
+
+
 def hierarchical_all_reduce(tensor, name=None):
-    """Local (intra-host) RCCL reduce -> cross-host CPU all-reduce among
-    local masters -> local RCCL broadcast (reference
-    ops/gpu/collective.cpp:106-158). On one host this degrades to a plain
-    RCCL all-reduce."""
+    """Local (intra-host) reduce to the local master -> cross-host
+    all-reduce among masters -> local broadcast (reference
+    ops/gpu/collective.cpp:106-158 ScheduledHierarchicalNcclAllReduce).
+    On one host this degrades to a plain all-reduce. With a live
+    torch.distributed process group the three hops run on sub-groups
+    (RCCL on GPUs); otherwise the C++ engine's local/cross/local
+    strategies carry it."""
+    import torch.distributed as dist
+
     _ensure_init()
     if _core.host_count() <= 1:
         return all_reduce(tensor, name=name)
-    if tensor.is_cuda:
-        # local reduce over the node group would need a sub-group; for the
-        # common single-host-per-8-GPU case we fall through to the staged
-        # cross path only when the node count > 1
-        tensor = cpu_staged_all_reduce(tensor, name=name)
+    if dist.is_available() and dist.is_initialized():
+        g = _hierarchical_groups()
+        dist.reduce(tensor, dst=g["local_master"], group=g["local"])
+        if g["is_master"]:
+            dist.all_reduce(tensor, group=g["cross"])
+        dist.broadcast(tensor, src=g["local_master"], group=g["local"])
         return tensor
+    if tensor.is_cuda:
+        # no process group: stage through the host into the C++ engine
+        # (the reference's CrossAllReduceGpu D2H path)
+        return cpu_staged_all_reduce(tensor, name=name)
     t = tensor.contiguous()
     _core.local_reduce(t.data_ptr(), t.data_ptr(), t.numel(),
                        core_dtype(t.dtype), core_op("sum"),

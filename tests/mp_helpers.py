@@ -555,3 +555,22 @@ def multihost_body(rank, np):
     kf.barrier()
     kf.finalize()
     return out
+
+
+def hier_subgroup_body(rank, np):
+    import torch
+    import torch.distributed as dist
+    import kungfu_amd as kf
+    from kungfu_amd.ops import hierarchical_all_reduce
+
+    kf.init()  # gloo process group via KUNGFU_TORCH_BACKEND
+    assert dist.is_initialized()
+    t = torch.full((1000,), float(rank + 1))
+    hierarchical_all_reduce(t, name="hsub")
+    out = float(t[0])
+    # run twice: groups are cached per cluster version
+    t2 = torch.ones(10)
+    hierarchical_all_reduce(t2, name="hsub2")
+    out2 = float(t2[0])
+    kf.finalize()
+    return (out, out2)
