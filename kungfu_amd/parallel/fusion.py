@@ -20,6 +20,11 @@ SURVEY.md §2.6 items 6/11 and §2.5 NCCLScheduler):
 
 The same reducer works for CPU tensors through the C++ collective engine
 (chunked graph strategies over TCP/Unix sockets) for plumbing mode.
+
+Overlap mode assumes ONE backward pass per optimizer step (the standard
+loop). Gradient-accumulation schedules (several backwards between steps)
+must use overlap=False; an accumulation into an already-reduced bucket
+raises instead of silently dropping peer contributions.
 """
 import torch
 
@@ -119,6 +124,14 @@ class GradBucketReducer:
 
     def _on_grad_ready(self, p):
         b = self.bucket_of[p]
+        if b.launched:
+            # a second backward pass accumulated into a bucket whose
+            # all-reduce already went out: results would silently drop the
+            # peer contributions of the first pass
+            raise RuntimeError(
+                "gradient accumulated after its bucket's all-reduce was "
+                "launched; use overlap=False for multi-backward "
+                "(gradient accumulation) steps")
         b.ready += 1
         if b.ready == len(b.params):
             self._drain()
