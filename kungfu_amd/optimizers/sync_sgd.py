@@ -67,6 +67,23 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
                 weight_decay=g.get("weight_decay", 0.0), grad_scale=scale,
                 nesterov=g.get("nesterov", False))
 
+    def state_dict(self):
+        sd = {"inner": self.optimizer.state_dict()}
+        if self.fused_step:
+            # momentum lives in the fused flat buffers, not in torch state
+            sd["fused_momentum"] = [b.momentum for b in
+                                    self.reducer.buckets]
+        return sd
+
+    def load_state_dict(self, sd):
+        if "inner" not in sd:  # plain torch-style dict (older checkpoints)
+            self.optimizer.load_state_dict(sd)
+            return
+        self.optimizer.load_state_dict(sd["inner"])
+        if self.fused_step and "fused_momentum" in sd:
+            for b, m in zip(self.reducer.buckets, sd["fused_momentum"]):
+                b.momentum.copy_(m.to(b.momentum.device))
+
     def _step(self):
         if self.fused_step:
             # skip the separate grad-average pass: 1/N is folded into the

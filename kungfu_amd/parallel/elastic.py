@@ -63,8 +63,8 @@ def save_checkpoint(path, model, optimizer=None, step=0, extra=None):
         "extra": extra or {},
     }
     if optimizer is not None:
-        inner = getattr(optimizer, "optimizer", optimizer)
-        state["optimizer"] = inner.state_dict()
+        # wrappers define state_dict (covering fused momentum buffers)
+        state["optimizer"] = optimizer.state_dict()
     torch.save(state, "%s.rank%d" % (path, kf.rank()))
 
 
@@ -78,6 +78,5 @@ def load_checkpoint(path, model, optimizer=None, map_location="cpu"):
     state = torch.load(f, map_location=map_location)
     model.load_state_dict(state["model"])
     if optimizer is not None and "optimizer" in state:
-        inner = getattr(optimizer, "optimizer", optimizer)
-        inner.load_state_dict(state["optimizer"])
+        optimizer.load_state_dict(state["optimizer"])
     return state.get("step", 0), state.get("extra", {})

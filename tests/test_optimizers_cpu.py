@@ -207,3 +207,34 @@ def sma_gns_body(rank, np):
 
 def test_sma_with_gns_probe(port_block):
     assert all(spawn_cluster(sma_gns_body, 2, port_block))
+
+
+def grad_accum_body(rank, np):
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import broadcast_model
+    from kungfu_amd.optimizers import SynchronousSGDOptimizer
+
+    kf.init(with_torch=False)
+    model = _make_model(seed=900 + rank)
+    broadcast_model(model)
+    # overlap=False supports gradient accumulation (2 backwards per step)
+    opt = SynchronousSGDOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.1), overlap=False)
+    torch.manual_seed(77 + rank)
+    for _ in range(2):
+        opt.zero_grad()
+        for _ in range(2):  # micro-batches
+            x = torch.randn(4, 32)
+            y = torch.randint(0, 4, (4,))
+            torch.nn.functional.cross_entropy(model(x), y).backward()
+        opt.step()
+    w = model.fc.weight.detach().flatten()
+    digest = [round(float(v), 6) for v in w[:5]]
+    kf.finalize()
+    return digest
+
+
+def test_grad_accumulation_no_overlap(port_block):
+    a, b = spawn_cluster(grad_accum_body, 2, port_block)
+    assert a == b  # replicas identical after accumulated synced steps
