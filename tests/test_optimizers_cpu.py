@@ -238,3 +238,33 @@ def grad_accum_body(rank, np):
 def test_grad_accumulation_no_overlap(port_block):
     a, b = spawn_cluster(grad_accum_body, 2, port_block)
     assert a == b  # replicas identical after accumulated synced steps
+
+
+def gns_zero_noise_body(rank, np):
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.optimizers import MonitorGradientNoiseScaleOptimizer
+
+    kf.init(with_torch=False)
+    model = _make_model(seed=950)      # identical init
+    opt = MonitorGradientNoiseScaleOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.0), device_batch_size=8)
+    torch.manual_seed(11)              # IDENTICAL data on every rank
+    for _ in range(2):
+        x = torch.randn(8, 32)
+        y = torch.randint(0, 4, (8,))
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        opt.step()
+    gns = opt.noise_scale
+    kf.finalize()
+    return gns
+
+
+def test_gns_zero_when_ranks_agree(port_block):
+    """Identical data on all ranks => local grad == averaged grad =>
+    the noise-scale estimator's S term is ~0 (formula sanity check,
+    reference ops/monitor.py:6-18)."""
+    results = spawn_cluster(gns_zero_noise_body, 2, port_block)
+    for gns in results:
+        assert abs(gns) < 1e-3, gns
