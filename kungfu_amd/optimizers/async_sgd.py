@@ -46,7 +46,9 @@ class PairAveragingOptimizer(KungFuOptimizer):
     def __init__(self, optimizer, fuse_requests=True, prefetch=False,
                  name=_MODEL_KEY, peer_selection="random",
                  exchange="store"):
-        """peer_selection: 'random' (AD-PSGD gossip) or 'roundrobin'
+        """fuse_requests: accepted for reference-API compatibility; the
+        model is always exchanged as one fused flat buffer here.
+        peer_selection: 'random' (AD-PSGD gossip) or 'roundrobin'
         (reference GetNeighbour/RoundRobin ops, ops/cpu/topology.cpp).
         exchange: 'store' (asymmetric pulls from the P2P blob store, the
         reference's model) or 'rccl' (symmetric tournament gossip with

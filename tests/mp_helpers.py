@@ -574,3 +574,25 @@ def hier_subgroup_body(rank, np):
     out2 = float(t2[0])
     kf.finalize()
     return (out, out2)
+
+
+def run_launcher_graceful(cmd, cwd, env, timeout):
+    """Run a kungfu-run invocation; on timeout SIGTERM it first (its signal
+    handler kills the worker process groups) before killing, so a failing
+    test cannot leak orphan workers that poison later tests' ports."""
+    import signal
+    import subprocess
+
+    proc = subprocess.Popen(cmd, cwd=cwd, env=env, stdout=subprocess.PIPE,
+                            stderr=subprocess.PIPE, text=True)
+    try:
+        out, err = proc.communicate(timeout=timeout)
+    except subprocess.TimeoutExpired:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            out, err = proc.communicate(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            out, err = proc.communicate()
+        return 124, out or "", err or ""
+    return proc.returncode, out, err

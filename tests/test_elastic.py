@@ -10,11 +10,17 @@ ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
 def run_watch(args, timeout=240):
+    import collections
+
+    from mp_helpers import run_launcher_graceful
+
     env = dict(os.environ)
     env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
-    return subprocess.run(
-        [sys.executable, "-m", "kungfu_amd.run"] + args,
-        cwd=ROOT, env=env, capture_output=True, text=True, timeout=timeout)
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run"] + args, ROOT, env,
+        timeout)
+    R = collections.namedtuple("R", "returncode stdout stderr")
+    return R(rc, out, err)
 
 
 def _plain(s):

@@ -13,14 +13,20 @@ def test_auto_recover_restarts_after_crash(port_block, tmp_path):
     env = dict(os.environ)
     env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
     ckpt = str(tmp_path / "ckpt.pt")
-    r = subprocess.run(
+    import collections
+
+    from mp_helpers import run_launcher_graceful
+
+    rc, stdout, stderr = run_launcher_graceful(
         [sys.executable, "-m", "kungfu_amd.run",
          "-np", "2", "-port", str(port_block), "-port-range",
          str(port_block + 1), "-auto-recover", "3s",
          "-monitor-port", str(port_block + 60),
          sys.executable, "examples/failure_recovery_trainer.py",
          "--n-epochs", "4", "--crash-at-epoch", "2", "--ckpt", ckpt],
-        cwd=ROOT, env=env, capture_output=True, text=True, timeout=300)
+        ROOT, env, 300)
+    r = collections.namedtuple("R", "returncode stdout stderr")(
+        rc, stdout, stderr)
     out = re.sub(r"\x1b\[[0-9;]*m", "", r.stdout)
     assert "CRASHING rank 0 now" in out
     assert "failure detected" in out

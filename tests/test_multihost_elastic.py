@@ -5,11 +5,33 @@ multi-runner stage-notification path (reference: cluster-in-docker tests +
 watch.go across hosts)."""
 import os
 import re
+import signal
 import subprocess
 import sys
 import time
 
+from mp_helpers import run_launcher_graceful  # noqa: F401
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _communicate_all(procs, deadline):
+    """communicate() with graceful SIGTERM on timeout so runners clean up
+    their worker process groups (no orphan port squatters)."""
+    outs = []
+    for p in procs:
+        try:
+            out, err = p.communicate(timeout=max(5, deadline - time.time()))
+            outs.append((p.returncode, _plain(out), _plain(err)))
+        except subprocess.TimeoutExpired:
+            p.send_signal(signal.SIGTERM)
+            try:
+                out, err = p.communicate(timeout=15)
+            except subprocess.TimeoutExpired:
+                p.kill()
+                out, err = p.communicate()
+            outs.append((124, _plain(out or ""), _plain(err or "")))
+    return outs
 
 
 def _plain(s):
@@ -46,11 +68,7 @@ def test_two_runners_elastic_grow(port_block):
                  "--schedule", "3:6", "--max-step", "7"],
                 cwd=ROOT, env=env, stdout=subprocess.PIPE,
                 stderr=subprocess.PIPE, text=True))
-        outs = []
-        deadline = time.time() + 240
-        for p in procs:
-            out, err = p.communicate(timeout=max(10, deadline - time.time()))
-            outs.append((p.returncode, _plain(out), _plain(err)))
+        outs = _communicate_all(procs, time.time() + 240)
         for rc, out, err in outs:
             assert rc == 0, out + err
         all_out = "\n".join(o for _, o, _ in outs)
@@ -87,10 +105,7 @@ def test_two_runners_auto_recover(port_block):
              "--n-epochs", "4", "--crash-at-epoch", "2", "--ckpt", ckpt],
             cwd=ROOT, env=env, stdout=subprocess.PIPE,
             stderr=subprocess.PIPE, text=True))
-    outs = []
-    for p in procs:
-        out, err = p.communicate(timeout=240)
-        outs.append((p.returncode, _plain(out), _plain(err)))
+    outs = _communicate_all(procs, time.time() + 240)
     all_out = "\n".join(o for _, o, _ in outs)
     for rc, out, err in outs:
         assert rc == 0, all_out + err
