@@ -77,10 +77,6 @@ def main():
     kf.finalize()
 
 
-if __name__ == "__main__":
-    main()
-
-
 def replace_cluster(new_workers):
     """Test helper: PUT an arbitrary worker list to the config server
     (exercises resizes the schedule API cannot express, e.g. removing
@@ -98,3 +94,7 @@ def replace_cluster(new_workers):
     }).encode()
     req = urllib.request.Request(url + "/config", data=body, method="PUT")
     urllib.request.urlopen(req, timeout=5)
+
+
+if __name__ == "__main__":
+    main()

@@ -11,7 +11,8 @@ from mp_helpers import (hier_subgroup_body, multihost_body,
 
 @pytest.mark.parametrize("strategy", ["AUTO", "BINARY_TREE_STAR", "RING",
                                       "MULTI_BINARY_TREE_STAR",
-                                      "MULTI_STAR", "TREE"])
+                                      "MULTI_STAR", "TREE", "CLIQUE",
+                                      "BINARY_TREE", "STAR"])
 def test_two_hosts_collectives(strategy, port_block):
     results = spawn_multihost(multihost_body, 4, port_block, strategy,
                               hosts=2)

@@ -78,3 +78,13 @@ def test_all_dtypes_reduce(port_block):
     from mp_helpers import dtype_sweep_body
 
     assert spawn_cluster(dtype_sweep_body, 2, port_block) == [True, True]
+
+
+def test_allreduce_np8(port_block):
+    """Full-node-shaped CPU cluster (8 workers, the driver's max)."""
+    results = spawn_cluster(allreduce_body, 8, port_block, "RING",
+                            timeout=180)
+    expect = sum(range(1, 9))
+    for r in results:
+        assert r["small_sum"] == pytest.approx(expect)
+        assert r["gathered"] == list(range(8))
