@@ -596,3 +596,29 @@ def run_launcher_graceful(cmd, cwd, env, timeout):
             out, err = proc.communicate()
         return 124, out or "", err or ""
     return proc.returncode, out, err
+
+
+def async_stress_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    rng = np_.random.RandomState(0)
+    sizes = [int(s) for s in rng.randint(1, 400_000, size=24)]
+    for round_ in range(3):
+        bufs = [np_.full(n, float(rank + 1), dtype=np_.float32)
+                for n in sizes]
+        handles = [
+            _core.all_reduce_async(b.ctypes.data, b.ctypes.data, b.size,
+                                   10, 0, "st%d" % i)
+            for i, b in enumerate(bufs)
+        ]
+        kf.barrier()  # a collective interleaved with the async storm
+        for h in handles:
+            _core.wait_handle(h)
+        expect = float(sum(range(1, np + 1)))
+        for b in bufs:
+            assert float(b[0]) == expect and float(b[-1]) == expect
+    kf.finalize()
+    return True

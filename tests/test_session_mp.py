@@ -88,3 +88,12 @@ def test_allreduce_np8(port_block):
     for r in results:
         assert r["small_sum"] == pytest.approx(expect)
         assert r["gathered"] == list(range(8))
+
+
+def test_async_stress(port_block):
+    """24 concurrent async all-reduces x 3 rounds with barriers
+    interleaved: shakes the name-keyed rendezvous under concurrency."""
+    from mp_helpers import async_stress_body
+
+    assert spawn_cluster(async_stress_body, 3, port_block,
+                         timeout=180) == [True] * 3
