@@ -17,3 +17,70 @@ def test_pair_averaging_store_two_procs_one_gpu(port_block):
                                "CUDA_VISIBLE_DEVICES": "0"},
                         timeout=240)
     assert res == [True, True]
+
+
+def test_elastic_resize_with_gpu_pair_averaging(port_block):
+    """BASELINE configs 3+5 on hardware: elastic grow 2->3 mid-run while
+    the model lives on the GPU and peers gossip through the store
+    (workers share cuda:0; RCCL needs distinct devices so the S-SGD path
+    is covered by the driver's multi-GPU bench instead)."""
+    import os
+    import re
+    import subprocess
+    import sys
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = r"""
+import os
+import torch
+import kungfu_amd as kf
+from kungfu_amd.models import SLP
+from kungfu_amd.optimizers import PairAveragingOptimizer
+
+kf.init(with_torch=False)
+torch.manual_seed(0)
+model = SLP(in_features=32, classes=4).to("cuda")
+opt = PairAveragingOptimizer(torch.optim.SGD(model.parameters(), lr=0.01))
+step = kf.all_reduce_int_max(0)
+print("JOIN rank=%d size=%d" % (kf.rank(), kf.size()), flush=True)
+while step < 6:
+    x = torch.randn(8, 1, 4, 8, device="cuda")
+    y = torch.randint(0, 4, (8,), device="cuda")
+    opt.zero_grad()
+    torch.nn.functional.cross_entropy(model(x), y).backward()
+    opt.step()
+    step += 1
+    if step == 2 and kf.rank() == 0:
+        kf.propose_new_size(3)
+    changed, detached = kf.resize()
+    if detached:
+        break
+    if changed:
+        step = kf.all_reduce_int_max(step)
+        opt = PairAveragingOptimizer(
+            torch.optim.SGD(model.parameters(), lr=0.01))
+        print("RESIZED size=%d" % kf.size(), flush=True)
+torch.cuda.synchronize()
+if not kf.detached():
+    print("DONE rank=%d size=%d step=%d" % (kf.rank(), kf.size(), step),
+          flush=True)
+kf.finalize()
+"""
+    env = dict(os.environ)
+    env.update({
+        "PYTHONPATH": ROOT + os.pathsep + env.get("PYTHONPATH", ""),
+        "HIP_VISIBLE_DEVICES": "0",
+        "CUDA_VISIBLE_DEVICES": "0",
+    })
+    from mp_helpers import run_launcher_graceful
+
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "2", "-w",
+         "-port", str(port_block), "-port-range", str(port_block + 2),
+         "-builtin-config-port", str(port_block + 1),
+         sys.executable, "-c", script], ROOT, env, 200)
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
+    assert rc == 0, plain + err
+    done = [ln for ln in plain.splitlines() if "DONE" in ln]
+    assert len(done) == 3 and all("size=3 step=6" in ln for ln in done), \
+        plain
