@@ -225,11 +225,19 @@ class Runner:
         except ProcessLookupError:
             pass
         try:
-            proc.popen.wait(timeout=10)
+            # short escalation: a worker blocked inside a collective never
+            # handles SIGTERM (GIL released in C++), and a slow teardown
+            # here can outlive a supervisor's own kill window, leaking
+            # port-squatting orphans
+            proc.popen.wait(timeout=3)
         except subprocess.TimeoutExpired:
             try:
                 os.killpg(proc.popen.pid, signal.SIGKILL)
             except ProcessLookupError:
+                pass
+            try:
+                proc.popen.wait(timeout=5)
+            except subprocess.TimeoutExpired:
                 pass
 
     # ---- config server ----

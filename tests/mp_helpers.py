@@ -590,7 +590,7 @@ def run_launcher_graceful(cmd, cwd, env, timeout):
     except subprocess.TimeoutExpired:
         proc.send_signal(signal.SIGTERM)
         try:
-            out, err = proc.communicate(timeout=15)
+            out, err = proc.communicate(timeout=40)
         except subprocess.TimeoutExpired:
             proc.kill()
             out, err = proc.communicate()

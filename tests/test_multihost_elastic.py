@@ -26,7 +26,7 @@ def _communicate_all(procs, deadline):
         except subprocess.TimeoutExpired:
             p.send_signal(signal.SIGTERM)
             try:
-                out, err = p.communicate(timeout=15)
+                out, err = p.communicate(timeout=40)
             except subprocess.TimeoutExpired:
                 p.kill()
                 out, err = p.communicate()
