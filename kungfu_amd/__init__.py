@@ -38,7 +38,7 @@ def _synthesize_env():
     if world <= 1:
         return  # single-process fallback inside _core
     rank = int(os.environ["RANK"])
-    base = int(os.environ.get("KUNGFU_PORT_BASE", "33100"))
+    base = int(os.environ.get("KUNGFU_PORT_BASE", "23100"))
     host = "127.0.0.1"
     peers = ",".join("%s:%d" % (host, base + i) for i in range(world))
     os.environ["KUNGFU_SELF_SPEC"] = "%s:%d" % (host, base + rank)

@@ -7,7 +7,10 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import random as _random
 
-_PORT_COUNTER = [_random.randrange(20000, 60000, 64)]
+# Listen ports must stay BELOW the kernel's ephemeral client-port range
+# (32768-60999 here): an in-flight outgoing socket can otherwise squat a
+# randomly-chosen listen port and fail a worker's bind.
+_PORT_COUNTER = [_random.randrange(20000, 30000, 64)]
 
 
 def _ancestors():

@@ -81,16 +81,20 @@ def spawn_cluster(fn, np, base, strategy="AUTO", timeout=90, extra=None,
                           "fresh port block", file=sys_mod.stderr)
                     for p in procs:
                         p.terminate()
-                    return spawn_cluster(fn, np, base + 8192, strategy,
-                                         timeout, extra, _retry=False)
+                    return spawn_cluster(fn, np,
+                                         20000 + (base + 4096) % 9000,
+                                         strategy, timeout, extra,
+                                         _retry=False)
                 raise AssertionError("cluster timed out; results so far: %r"
                                      % (results,))
             if (status == "err" and _retry
                     and "Address already in use" in str(payload)):
                 for p in procs:
                     p.terminate()
-                return spawn_cluster(fn, np, base + 8192, strategy,
-                                     timeout, extra, _retry=False)
+                return spawn_cluster(fn, np,
+                                     20000 + (base + 4096) % 9000,
+                                     strategy, timeout, extra,
+                                     _retry=False)
             assert status == "ok", "rank %d failed: %s" % (i, payload)
             results[i] = payload
     finally:
@@ -501,7 +505,8 @@ def run_worker_mh(fn, i, np, base, strategy, q, hosts):
         q.put((i, "err", "%s\n%s" % (e, traceback.format_exc())))
 
 
-def spawn_multihost(fn, np, base, strategy="AUTO", hosts=2, timeout=120):
+def spawn_multihost(fn, np, base, strategy="AUTO", hosts=2, timeout=120,
+                    _retry=True):
     import multiprocessing as mp
 
     ctx = mp.get_context("spawn")
@@ -515,6 +520,14 @@ def spawn_multihost(fn, np, base, strategy="AUTO", hosts=2, timeout=120):
     try:
         for _ in range(np):
             i, status, payload = q.get(timeout=timeout)
+            if (status == "err" and _retry
+                    and "Address already in use" in str(payload)):
+                for p in procs:
+                    p.terminate()
+                return spawn_multihost(fn, np,
+                                       20000 + (base + 4096) % 9000,
+                                       strategy, hosts, timeout,
+                                       _retry=False)
             assert status == "ok", "rank %d failed: %s" % (i, payload)
             results[i] = payload
     finally:
