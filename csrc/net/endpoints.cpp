@@ -25,6 +25,25 @@ std::shared_ptr<CollectiveEndpoint::Slot> CollectiveEndpoint::slot(
 
 void CollectiveEndpoint::on_frame(const PeerID &src, Frame &f)
 {
+    if (f.flags & msgflag::ShmRef) {
+        // colocated shm data path: payload is a /dev/shm file reference;
+        // swap it for the file contents and unlink (single consumer)
+        std::string path((const char *)f.data.data(), f.data.size());
+        FILE *fp = std::fopen(path.c_str(), "rb");
+        if (!fp)
+            throw std::runtime_error("shm chunk missing: " + path);
+        std::fseek(fp, 0, SEEK_END);
+        long n = std::ftell(fp);
+        std::fseek(fp, 0, SEEK_SET);
+        std::vector<uint8_t> buf((size_t)n);
+        size_t got = std::fread(buf.data(), 1, (size_t)n, fp);
+        std::fclose(fp);
+        ::unlink(path.c_str());
+        if (got != (size_t)n)
+            throw std::runtime_error("short shm chunk read: " + path);
+        f.data = std::move(buf);
+        f.flags &= ~msgflag::ShmRef;
+    }
     auto s = slot(src, f.name);
     std::lock_guard<std::mutex> lk(s->mu);
     if (s->dst && !s->filled) {

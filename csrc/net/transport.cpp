@@ -10,6 +10,7 @@
 
 #include <chrono>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <stdexcept>
 
@@ -366,6 +367,32 @@ void Client::send(const PeerID &remote, ConnType type,
                   const std::string &name, uint32_t flags, const void *data,
                   size_t len)
 {
+    // Optional /dev/shm data path for colocated collective payloads
+    // (KUNGFU_SHM_COLLECTIVES=1): the socket carries only a file
+    // reference; the write completes before the frame is sent, so the
+    // receiver's read-after-frame ordering is safe. The receiver unlinks
+    // after consuming (endpoints.cpp).
+    std::string shm_payload;
+    if (shm_collectives_ && type == ConnType::Collective &&
+        len >= (64u << 10) && remote.ipv4 == self_.ipv4 &&
+        (flags & msgflag::ShmRef) == 0) {
+        shm_payload = "/dev/shm/kfc-" + std::to_string(self_.key()) + "-" +
+                      std::to_string(shm_seq_.fetch_add(1));
+        FILE *f = std::fopen(shm_payload.c_str(), "wb");
+        bool ok = f != nullptr;
+        if (f) {
+            ok = std::fwrite(data, 1, len, f) == len;
+            std::fclose(f);
+        }
+        if (ok) {
+            data = shm_payload.data();
+            len = shm_payload.size();
+            flags |= msgflag::ShmRef;
+        } else {
+            if (f) ::unlink(shm_payload.c_str());
+            shm_payload.clear();  // fall back to inline payload
+        }
+    }
     for (int attempt = 0; attempt < 2; ++attempt) {
         auto conn = get_conn(remote, type);
         if (conn->send_frame(name, flags, data, len)) {

@@ -100,7 +100,11 @@ class Server {
 // Outgoing connection pool: one cached conn per (remote, type).
 class Client {
   public:
-    explicit Client(const PeerID &self) : self_(self) {}
+    explicit Client(const PeerID &self) : self_(self)
+    {
+        const char *e = std::getenv("KUNGFU_SHM_COLLECTIVES");
+        shm_collectives_ = e && *e && std::string(e) != "0";
+    }
     ~Client() { reset({}, 0); }
 
     void set_token(uint32_t token) { token_ = token; }
@@ -123,6 +127,8 @@ class Client {
                                    int connect_timeout_ms = 20000);
 
     PeerID self_;
+    bool shm_collectives_ = false;
+    std::atomic<uint64_t> shm_seq_{0};
     std::atomic<uint32_t> token_{0};
     mutable std::mutex mu_;
     std::map<std::pair<uint64_t, uint8_t>, std::shared_ptr<Conn>> pool_;

@@ -97,3 +97,16 @@ def test_async_stress(port_block):
 
     assert spawn_cluster(async_stress_body, 3, port_block,
                          timeout=180) == [True] * 3
+
+
+@pytest.mark.parametrize("strategy", ["STAR", "RING"])
+def test_shm_collective_path(strategy, port_block):
+    """Opt-in /dev/shm data path for colocated collective chunks
+    (KUNGFU_SHM_COLLECTIVES=1): payloads >= 64 KiB ride tmpfs files, the
+    socket carries only references."""
+    results = spawn_cluster(allreduce_body, 4, port_block, strategy,
+                            extra={"KUNGFU_SHM_COLLECTIVES": "1"})
+    expect_sum = sum(range(1, 5))
+    for r in results:
+        assert r["small_sum"] == pytest.approx(expect_sum)
+        assert r["gathered"] == list(range(4))
