@@ -635,3 +635,18 @@ def async_stress_body(rank, np):
             assert float(b[0]) == expect and float(b[-1]) == expect
     kf.finalize()
     return True
+
+
+def egress_rates_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+    from kungfu_amd.ops import egress_rates
+
+    kf.init(with_torch=False)
+    egress_rates()  # baseline snapshot
+    x = np_.ones(200_000, dtype=np_.float32)
+    _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "e")
+    rates = egress_rates()
+    kf.finalize()
+    return any(v > 0 for v in rates.values())

@@ -62,3 +62,31 @@ def test_fakemodel_sizes():
     assert total_params("slp-mnist") == 28 * 28 * 10 + 10
     assert 108e6 < total_params("bert") < 135e6
     assert 135e6 < total_params("vgg16-imagenet") < 140e6
+
+
+def test_adaptive_example(port_block):
+    import os
+    import subprocess
+    import sys
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    from mp_helpers import run_launcher_graceful
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "3", "-port",
+         str(port_block), "-port-range", str(port_block + 1),
+         "-strategy", "STAR",
+         sys.executable, "examples/adaptive_trainer.py", "--steps", "8",
+         "--mst"], ROOT, env, 180)
+    assert rc == 0, out + err
+    assert out.count("ADAPT-DONE") == 3
+    assert "MST parent array" in out
+    assert "strategy 0:" in out  # stats printed
+
+
+def test_egress_rates_helper(port_block):
+    from mp_helpers import egress_rates_body
+
+    assert all(spawn_cluster(egress_rates_body, 2, port_block))
