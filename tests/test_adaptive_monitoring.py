@@ -1,8 +1,8 @@
 """Adaptive strategy switching, metrics endpoint, elastic sampler,
 policy runner (reference: adaptiveStrategies.go, monitor/server.go,
 datasets/adaptor.py, policy/)."""
-from mp_helpers import (adaptive_body, metrics_body, sampler_body,
-                        spawn_cluster)
+from mp_helpers import (adaptive_body, metrics_body, retry_flaky,
+                        sampler_body, spawn_cluster)
 
 
 def test_interference_vote_switches_strategy(port_block):
@@ -64,6 +64,7 @@ def test_fakemodel_sizes():
     assert 135e6 < total_params("vgg16-imagenet") < 140e6
 
 
+@retry_flaky
 def test_adaptive_example(port_block):
     import os
     import subprocess

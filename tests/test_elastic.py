@@ -6,6 +6,8 @@ import re
 import subprocess
 import sys
 
+from mp_helpers import retry_flaky
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
@@ -27,6 +29,7 @@ def _plain(s):
     return re.sub(r"\x1b\[[0-9;]*m", "", s)
 
 
+@retry_flaky
 def test_elastic_grow_2_to_3(port_block):
     r = run_watch([
         "-np", "2", "-w", "-port", str(port_block), "-port-range",
@@ -44,6 +47,7 @@ def test_elastic_grow_2_to_3(port_block):
     assert all("size=3 step=8" in ln for ln in done)
 
 
+@retry_flaky
 def test_elastic_shrink_3_to_2(port_block):
     r = run_watch([
         "-np", "3", "-w", "-port", str(port_block), "-port-range",
@@ -59,6 +63,7 @@ def test_elastic_shrink_3_to_2(port_block):
     assert all("size=2 step=6" in ln for ln in done)
 
 
+@retry_flaky
 def test_elastic_grow_and_shrink(port_block):
     r = run_watch([
         "-np", "2", "-w", "-port", str(port_block), "-port-range",
@@ -74,6 +79,7 @@ def test_elastic_grow_and_shrink(port_block):
     assert out.count("DETACHED") == 2
 
 
+@retry_flaky
 def test_elastic_churn(port_block):
     """Repeated grow/shrink churn: 2 -> 4 -> 2 -> 3 -> 2 in one run."""
     r = run_watch([
@@ -90,6 +96,7 @@ def test_elastic_churn(port_block):
     assert out.count("RESIZED") >= 6  # at least surviving workers log each
 
 
+@retry_flaky
 def test_elastic_remove_rank0(port_block):
     """Removing rank 0 mid-run: the old rank 0 notifies the runners, then
     detaches; the old rank 1 becomes the new rank 0 and training continues

@@ -5,10 +5,12 @@ import os
 import re
 import subprocess
 import sys
+from mp_helpers import retry_flaky
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+@retry_flaky
 def test_auto_recover_restarts_after_crash(port_block, tmp_path):
     env = dict(os.environ)
     env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")

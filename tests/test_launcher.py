@@ -3,6 +3,7 @@ run-integration-tests.sh / run-train-tests.sh under the real launcher)."""
 import os
 import subprocess
 import sys
+from mp_helpers import retry_flaky
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
@@ -21,6 +22,7 @@ def run_launcher(args, timeout=180):
     return R(rc, out, err)
 
 
+@retry_flaky
 def test_kungfu_run_np2_mnist_slp(port_block):
     r = run_launcher([
         "-np", "2", "-port", str(port_block), "-port-range",
@@ -31,6 +33,7 @@ def test_kungfu_run_np2_mnist_slp(port_block):
     assert r.stdout.count("FINAL") == 2
 
 
+@retry_flaky
 def test_kungfu_run_env_protocol(port_block):
     script = ("import kungfu_amd as kf, os; kf.init(with_torch=False); "
               "print('R', kf.rank(), kf.size(), kf.local_rank(), "
@@ -49,6 +52,7 @@ def test_kungfu_run_env_protocol(port_block):
     assert lines == ["R 0 3 0 0", "R 1 3 1 1", "R 2 3 2 2"], repr(plain)
 
 
+@retry_flaky
 def test_kungfu_run_propagates_failure(port_block):
     r = run_launcher([
         "-np", "2", "-port", str(port_block), "-port-range",

@@ -11,6 +11,7 @@ import sys
 import time
 
 from mp_helpers import run_launcher_graceful  # noqa: F401
+from mp_helpers import retry_flaky
 
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
@@ -38,6 +39,7 @@ def _plain(s):
     return re.sub(r"\x1b\[[0-9;]*m", "", s)
 
 
+@retry_flaky
 def test_two_runners_elastic_grow(port_block):
     from kungfu_amd.launcher.configserver import make_server
 
@@ -84,6 +86,7 @@ def test_two_runners_elastic_grow(port_block):
         srv.shutdown()
 
 
+@retry_flaky
 def test_two_runners_auto_recover(port_block):
     """Cross-host failure recovery: a worker on host 2 crashes; host 1's
     monitor detects the stall (all heartbeats go to host 0 = runners[0]),

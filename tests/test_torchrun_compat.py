@@ -5,9 +5,12 @@ import re
 import subprocess
 import sys
 
+from mp_helpers import retry_flaky
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+@retry_flaky
 def test_bench_slp_under_torchrun_env(port_block):
     procs = []
     for rank in range(2):
@@ -38,6 +41,7 @@ def test_bench_slp_under_torchrun_env(port_block):
     assert '"n_gpus": 2' in jsons[0]
 
 
+@retry_flaky
 def test_elastic_trainer_class(port_block):
     env = dict(os.environ)
     env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
