@@ -107,16 +107,22 @@ def _maybe_init_torch_dist():
 
 
 def _reinit_torch_dist():
-    """Tear down and re-create the RCCL process group after a resize
-    (reference: ResetNcclHelper, ops/gpu/scheduler.cpp:43-72)."""
+    """Tear down and re-create the torch.distributed process group after a
+    resize (reference: ResetNcclHelper, ops/gpu/scheduler.cpp:43-72).
+
+    Gated on whether a group actually exists / could exist — NOT on CUDA
+    availability: a gloo group (KUNGFU_TORCH_BACKEND=gloo) must also be
+    destroyed and re-created, or survivors keep a stale world size while
+    joiners block in init_process_group."""
     global _torch_dist
     import torch
-
-    if not torch.cuda.is_available():
-        return
     import torch.distributed as dist
 
-    if dist.is_initialized():
+    had_group = dist.is_available() and dist.is_initialized()
+    if (not had_group and not torch.cuda.is_available()
+            and not os.environ.get("KUNGFU_TORCH_BACKEND")):
+        return  # pure CPU-plumbing mode: no process group in play
+    if had_group:
         dist.destroy_process_group()
     _torch_dist = False
     if _core.size() > 1:

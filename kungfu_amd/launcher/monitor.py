@@ -134,11 +134,19 @@ def monitored_run(runner, grace=GRACE_SECONDS):
     srv = start_monitor_server(state, runner.args.monitor_port,
                                runner.args.self_ip)
     world = len(runner.peers.split(","))
+    # Workers on EVERY host heartbeat the FIRST host's monitor
+    # (kungfu_amd.cmd._monitor_host), so trainend accumulates global ranks
+    # there and never arrives on other hosts: the first host must compare
+    # against the global world size (not its local worker count), and the
+    # other hosts complete only via local process exit.
+    first_ip = runner.runners.split(",")[0].rsplit(":", 1)[0]
+    is_first_host = first_ip == runner.args.self_ip
     try:
         attempt = 0
         while True:
-            state.down = False
-            state.begin.clear()
+            with state.lock:
+                state.down = False
+                state.begin.clear()
             peers_csv = runner.peers
             for spec in runner.local_specs(peers_csv):
                 runner.spawn(spec, peers_csv, runner.version)
@@ -158,8 +166,8 @@ def monitored_run(runner, grace=GRACE_SECONDS):
                             runner.procs.pop(spec, None)
                         code = max(code, rc)
                 with state.lock:
-                    done = len(state.trainend) >= len(
-                        runner.local_specs(peers_csv))
+                    done = (is_first_host and
+                            len(state.trainend) >= world)
                 if done or (all_exited and not live):
                     return code
                 if state.check_down(grace):
@@ -168,7 +176,8 @@ def monitored_run(runner, grace=GRACE_SECONDS):
                 time.sleep(0.5)
             if failed:
                 attempt += 1
-                locally_detected = state.other_down_epoch is None
+                with state.lock:
+                    locally_detected = state.other_down_epoch is None
                 min_epoch = state.min_epoch()
                 print("[kungfu-run] failure detected (attempt %d); "
                       "restarting from epoch %d" % (attempt, min_epoch),
@@ -183,8 +192,9 @@ def monitored_run(runner, grace=GRACE_SECONDS):
                     runner.kill(s)
                 runner.args.prog = _adjust_prog(runner.args.prog,
                                                 min_epoch)
-                state.trainend.clear()
-                state.other_down_epoch = None
+                with state.lock:
+                    state.trainend.clear()
+                    state.other_down_epoch = None
                 time.sleep(1.0)
     finally:
         srv.shutdown()

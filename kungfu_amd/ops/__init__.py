@@ -11,10 +11,25 @@ Routing (MI355X-native):
     (src/torch/ops/cuda/collective.cpp:31-54) kept for the cross-host hop
     and plumbing tests.
 """
+import collections
+import itertools
+
 import torch
 
 from kungfu_amd import _core, _ensure_init
 from kungfu_amd.utils.dtypes import core_dtype, core_op
+
+# Per-(kind, numel) sequence counters for auto-derived collective names:
+# two successive collectives on distinct same-sized tensors must not share
+# a rendezvous name or their payloads could interleave on the CPU engine.
+# Counters advance identically on every rank because collectives are (by
+# contract) called in the same program order on all ranks. Concurrent
+# collectives from MULTIPLE threads must pass explicit unique `name`s.
+_name_seq = collections.defaultdict(itertools.count)
+
+
+def _auto_name(kind, numel):
+    return "%s%d#%d" % (kind, numel, next(_name_seq[(kind, numel)]))
 
 
 def _dist():
@@ -62,7 +77,7 @@ def all_reduce(tensor, op="sum", name=None, average=False, async_op=False):
     t = tensor.contiguous()
     _core.all_reduce(t.data_ptr(), t.data_ptr(), t.numel(),
                      core_dtype(t.dtype), core_op(op),
-                     name or "t%d" % t.numel())
+                     name or _auto_name("ar", t.numel()))
     if t.data_ptr() != tensor.data_ptr():
         tensor.copy_(t)
     if average:
@@ -79,7 +94,8 @@ def broadcast(tensor, root=0, name=None):
         return tensor
     t = tensor.contiguous()
     _core.broadcast(t.data_ptr(), t.data_ptr(), t.numel(),
-                    core_dtype(t.dtype), name or "t%d" % t.numel(), root)
+                    core_dtype(t.dtype),
+                    name or _auto_name("bc", t.numel()), root)
     if t.data_ptr() != tensor.data_ptr():
         tensor.copy_(t)
     return tensor
@@ -101,7 +117,8 @@ def all_gather(tensor, name=None):
         return out
     t = tensor.contiguous()
     _core.all_gather(t.data_ptr(), out.data_ptr(), t.numel(),
-                     core_dtype(t.dtype), name or "t%d" % t.numel())
+                     core_dtype(t.dtype),
+                     name or _auto_name("ag", t.numel()))
     return out
 
 
@@ -117,7 +134,7 @@ def reduce(tensor, op="sum", name=None):
     t = tensor.contiguous()
     _core.reduce(t.data_ptr(), t.data_ptr(), t.numel(),
                  core_dtype(t.dtype), core_op(op),
-                 name or "t%d" % t.numel())
+                 name or _auto_name("rd", t.numel()))
     if t.data_ptr() != tensor.data_ptr():
         tensor.copy_(t)
     return tensor
@@ -132,7 +149,7 @@ def cpu_staged_all_reduce(tensor, op="sum", name=None):
     host = tensor.detach().to("cpu", non_blocking=False).contiguous()
     _core.all_reduce(host.data_ptr(), host.data_ptr(), host.numel(),
                      core_dtype(host.dtype), core_op(op),
-                     name or "h%d" % host.numel())
+                     name or _auto_name("h", host.numel()))
     tensor.copy_(host.to(tensor.device))
     return tensor
 

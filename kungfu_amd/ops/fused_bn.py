@@ -77,6 +77,8 @@ class _FusedBNFunction(torch.autograd.Function):
         x, a, save_mean, save_rstd, mask = ctx.saved_tensors
         M, C = ctx.MC
         s = _stream()
+        if dy.dtype != x.dtype:  # kernels require bf16 dy (matching x)
+            dy = dy.to(x.dtype)
         dy = dy.contiguous(memory_format=torch.channels_last)
         dev = x.device
         sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
@@ -111,9 +113,19 @@ class FusedBNReLU2d(torch.nn.Module):
         self.register_buffer("running_mean", torch.zeros(channels))
         self.register_buffer("running_var", torch.ones(channels))
 
+    def _params_f32(self):
+        # the kernels reinterpret these pointers as float32: after e.g.
+        # model.to(torch.bfloat16) the fused path must fall back to eager
+        # instead of reading bf16 bytes as f32 garbage
+        return (self.weight.dtype == torch.float32 and
+                self.bias.dtype == torch.float32 and
+                self.running_mean.dtype == torch.float32 and
+                self.running_var.dtype == torch.float32)
+
     def forward(self, x, residual=None):
         if _hip is not None and x.is_cuda and _nhwc_ok(x) and (
-                residual is None or _nhwc_ok(residual)):
+                residual is None or _nhwc_ok(residual)) and \
+                self._params_f32():
             return _FusedBNFunction.apply(
                 x, residual, self.weight, self.bias, self.running_mean,
                 self.running_var, self.momentum, self.eps, self.relu,

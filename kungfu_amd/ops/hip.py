@@ -50,22 +50,44 @@ def _check_cuda(*tensors):
 
 class FusionPlan:
     """Device-side chunk table for packing a list of same-dtype tensors into
-    (or out of) a flat fused buffer."""
+    (or out of) a flat fused buffer.
+
+    Holds references to the tensors (keeping their storage alive) and
+    revalidates their data_ptrs on every pack/unpack: `p.data` reassignment
+    (fused-step flat aliasing, elastic rebuilds, model.to(...)) would
+    otherwise leave the device-side chunk table pointing at freed or
+    unrelated memory. On mismatch the chunk table is rebuilt in place."""
 
     def __init__(self, tensors, offsets, dtype, chunk_elems=1 << 14):
         _require()
         _check_cuda(*tensors)
-        segs = [(t.data_ptr(), int(off), t.numel())
-                for t, off in zip(tensors, offsets)]
-        self._plan = _hip.FusionPlan(segs, hip_dtype(dtype), chunk_elems)
+        self._tensors = list(tensors)
+        self._offsets = [int(o) for o in offsets]
+        self._chunk_elems = chunk_elems
+        self._ptrs = [t.data_ptr() for t in self._tensors]
+        self._plan = self._build()
         self.dtype = dtype
+
+    def _build(self):
+        segs = [(t.data_ptr(), off, t.numel())
+                for t, off in zip(self._tensors, self._offsets)]
+        return _hip.FusionPlan(segs, hip_dtype(self._tensors[0].dtype),
+                               self._chunk_elems)
+
+    def _revalidate(self):
+        ptrs = [t.data_ptr() for t in self._tensors]
+        if ptrs != self._ptrs:
+            self._ptrs = ptrs
+            self._plan = self._build()
 
     def pack(self, fused):
         _check_cuda(fused)
+        self._revalidate()
         self._plan.pack(fused.data_ptr(), _stream())
 
     def unpack(self, fused, scale=1.0):
         _check_cuda(fused)
+        self._revalidate()
         self._plan.unpack(fused.data_ptr(), float(scale), _stream())
 
     @property

@@ -86,11 +86,13 @@ class GradBucketReducer:
         order = list(reversed(self.params))
         self.buckets = []
         cur = _Bucket(0)
-        esize = order[0].element_size()
-        cap = max(1, bucket_bytes // esize)
         for p in order:
+            # capacity in elements of THIS bucket's dtype (buckets are
+            # homogeneous; a dtype change forces a bucket break)
+            cap = max(1, bucket_bytes // p.element_size())
+            dtype_break = cur.params and cur.params[0].dtype != p.dtype
             aligned = (p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
-            if cur.numel > 0 and cur.numel + aligned > cap:
+            if cur.numel > 0 and (cur.numel + aligned > cap or dtype_break):
                 self.buckets.append(cur)
                 cur = _Bucket(len(self.buckets))
             cur.params.append(p)
@@ -221,8 +223,10 @@ class FlatParamGroup:
         if self.device.type == "cuda":
             from kungfu_amd.ops import hip as hip_ops
 
-            self._plan = hip_ops.FusionPlan(
-                [p.data for p in self.params], offsets, self.dtype)
+            # pass the live parameters (not `.data` snapshots) so the plan
+            # can revalidate data_ptrs after p.data reassignment
+            self._plan = hip_ops.FusionPlan(self.params, offsets,
+                                            self.dtype)
 
     def pack(self):
         if self._plan is not None:
