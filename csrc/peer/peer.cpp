@@ -248,6 +248,14 @@ int Peer::local_size() const
     return workers_.local_size_of(cfg_.self);
 }
 int Peer::host_count() const { return workers_.host_count(); }
+int Peer::host_rank() const
+{
+    const auto hs = workers_.hosts();
+    for (int i = 0; i < (int)hs.size(); ++i) {
+        if (hs[i] == cfg_.self.ipv4) return i;
+    }
+    return 0;
+}
 
 Session &Peer::session()
 {

@@ -61,6 +61,7 @@ class Peer {
     int local_rank() const;
     int local_size() const;
     int host_count() const;
+    int host_rank() const;  // index of this peer's host among hosts
     uint32_t version() const { return version_; }
     std::string uid() const { return cfg_.self.str(); }
     bool detached() const { return detached_; }

@@ -17,6 +17,7 @@
 #include <mutex>
 #include <tuple>
 
+#include "../core/control_api.h"
 #include "../core/graph.hpp"
 #include "../core/plan.hpp"
 #include "../peer/peer.hpp"
@@ -65,6 +66,46 @@ Peer &peer()
     if (!g_peer) throw std::runtime_error("kungfu not initialized");
     return *g_peer;
 }
+
+// ---- control-plane C API (consumed by kungfu_amd._rccl) ----
+// The RCCL layer bootstraps its communicators over THIS control plane
+// (uniqueId broadcast, scheduler order agreement) instead of a parallel
+// TCPStore rendezvous; see csrc/core/control_api.h.
+extern "C" {
+static Peer *cap_peer(void *ctx) { return static_cast<Peer *>(ctx); }
+static int cap_rank(void *c) { return cap_peer(c)->rank(); }
+static int cap_size(void *c) { return cap_peer(c)->size(); }
+static int cap_local_rank(void *c) { return cap_peer(c)->local_rank(); }
+static int cap_local_size(void *c) { return cap_peer(c)->local_size(); }
+static int cap_host_count(void *c) { return cap_peer(c)->host_count(); }
+static int cap_host_rank(void *c) { return cap_peer(c)->host_rank(); }
+static uint32_t cap_version(void *c) { return cap_peer(c)->version(); }
+static void cap_broadcast(void *c, void *buf, size_t len, int root,
+                          const char *name)
+{
+    Workspace w;
+    w.send = buf;
+    w.recv = buf;
+    w.count = len;
+    w.dt = DType::U8;
+    w.name = name;
+    cap_peer(c)->session().broadcast(w, root);
+}
+static void cap_local_broadcast(void *c, void *buf, size_t len,
+                                const char *name)
+{
+    Workspace w;
+    w.send = buf;
+    w.recv = buf;
+    w.count = len;
+    w.dt = DType::U8;
+    w.name = name;
+    cap_peer(c)->session().local_broadcast(w);
+}
+static void cap_barrier(void *c) { cap_peer(c)->session().barrier(); }
+}  // extern "C"
+
+kf_control_api g_control_api;
 
 // Runtime tracing (reference: stdtracer TRACE_SCOPE, KUNGFU_ENABLE_TRACE;
 // include/kungfu/utils/trace.hpp): env-gated event ring readable from
@@ -235,6 +276,26 @@ PYBIND11_MODULE(_core, m)
         g_peer.reset();
     });
     m.def("initialized", [] { return (bool)g_peer; });
+    // Capsule handing the control plane to the RCCL layer (_rccl) for
+    // communicator-id rendezvous and order agreement. Valid until
+    // finalize(); _rccl.finalize() must run first (kungfu_amd.finalize
+    // orders this).
+    m.def("control_api", [] {
+        if (!g_peer) throw std::runtime_error("kungfu not initialized");
+        g_control_api.api_version = KF_CONTROL_API_VERSION;
+        g_control_api.ctx = g_peer.get();
+        g_control_api.rank = cap_rank;
+        g_control_api.size = cap_size;
+        g_control_api.local_rank = cap_local_rank;
+        g_control_api.local_size = cap_local_size;
+        g_control_api.host_count = cap_host_count;
+        g_control_api.host_rank = cap_host_rank;
+        g_control_api.cluster_version = cap_version;
+        g_control_api.broadcast = cap_broadcast;
+        g_control_api.local_broadcast = cap_local_broadcast;
+        g_control_api.barrier = cap_barrier;
+        return py::capsule(&g_control_api, KF_CONTROL_API_CAPSULE);
+    });
 
     // ---- metadata ----
     m.def("rank", [] { return peer().rank(); });

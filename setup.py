@@ -39,37 +39,61 @@ HIP_SOURCES = [
     "csrc/hip/module_hip.cpp",
 ]
 
+RCCL_SOURCES = [
+    "csrc/rccl/rccl_layer.cpp",
+    "csrc/rccl/module_rccl.cpp",
+]
+
 HIP_ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 
 
-def build_hip_ext(out_dir):
-    """Compile the HIP extension with hipcc directly (gfx950 only)."""
+def _hipcc_ext(name, sources, out_dir, extra_link=()):
+    """Compile one hipcc extension in-tree (gfx950 only)."""
     import sysconfig
 
     py_inc = sysconfig.get_paths()["include"]
     pb_inc = pybind11.get_include()
-    out = os.path.join(out_dir, "_hip" + sysconfig.get_config_var("EXT_SUFFIX"))
+    out = os.path.join(out_dir, name + sysconfig.get_config_var("EXT_SUFFIX"))
     objs = []
-    os.makedirs(os.path.join(ROOT, "build", "hip"), exist_ok=True)
-    for src in HIP_SOURCES:
+    os.makedirs(os.path.join(ROOT, "build", name), exist_ok=True)
+    for src in sources:
         obj = os.path.join(
-            ROOT, "build", "hip",
+            ROOT, "build", name,
             os.path.basename(src).replace(".", "_") + ".o")
+        src_path = os.path.join(ROOT, src)
+        import glob
+
+        deps = [src_path] + glob.glob(os.path.join(ROOT, "csrc", "**",
+                                                   "*.h*"), recursive=True)
+        if (os.path.exists(obj) and
+                os.path.getmtime(obj) > max(map(os.path.getmtime, deps))):
+            objs.append(obj)
+            continue
         cmd = [
-            HIPCC, "-c", os.path.join(ROOT, src), "-o", obj,
+            HIPCC, "-c", src_path, "-o", obj,
             f"--offload-arch={HIP_ARCH}", "-O3", "-std=c++17", "-fPIC",
-            f"-I{py_inc}", f"-I{pb_inc}",
+            f"-I{py_inc}", f"-I{pb_inc}", "-I/opt/rocm/include",
         ]
         print("+", " ".join(cmd))
         subprocess.check_call(cmd)
         objs.append(obj)
     link = [HIPCC, "-shared", "-fPIC", "-o", out] + objs + [
         f"--offload-arch={HIP_ARCH}"
-    ]
+    ] + list(extra_link)
     print("+", " ".join(link))
     subprocess.check_call(link)
     return out
+
+
+def build_hip_ext(out_dir):
+    return _hipcc_ext("_hip", HIP_SOURCES, out_dir)
+
+
+def build_rccl_ext(out_dir):
+    return _hipcc_ext("_rccl", RCCL_SOURCES, out_dir,
+                      extra_link=["-L/opt/rocm/lib", "-lrccl",
+                                  "-Wl,-rpath,/opt/rocm/lib"])
 
 
 class BuildExt(build_ext):
@@ -79,8 +103,9 @@ class BuildExt(build_ext):
             if os.path.exists(HIPCC):
                 dest = os.path.join(ROOT, "kungfu_amd")
                 build_hip_ext(dest)
+                build_rccl_ext(dest)
             else:
-                print("hipcc not found; skipping kungfu_amd._hip")
+                print("hipcc not found; skipping kungfu_amd._hip/_rccl")
 
 
 core_ext = Extension(

@@ -674,3 +674,88 @@ def retry_flaky(fn):
             return fn(random.randrange(20000, 29000, 64), *a, **k)
 
     return wrapper
+
+
+# ---- native RCCL layer: ordered dispatcher + order agreement (CPU) ----
+
+def rccl_order_agreement_body(rank, np):
+    """Adversarial ordering (VERDICT round-1 item 6): every rank enqueues
+    the same named tasks in a DIFFERENT arrival order; the dispatcher must
+    release them in the agreed order on every rank, and scheduler_agree
+    must adopt rank 0's arrival order cluster-wide over the control plane
+    (reference nccl/scheduler.cpp:93-119 semantics)."""
+    import kungfu_amd as kf
+    from kungfu_amd import _rccl
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    rccl.init_cpu()
+    names = ["bkt0", "bkt1", "bkt2", "bkt3"]
+    rccl.scheduler_reset(names)
+
+    # round 1: per-rank adversarial arrival; release must be identity
+    arrivals = {0: [2, 0, 3, 1], 1: [3, 1, 0, 2], 2: [1, 3, 2, 0]}
+    arrival = arrivals.get(rank % 3, list(range(4)))
+    executed = []
+    for slot in arrival:
+        _rccl.start_task(0, names[slot],
+                         (lambda s=slot: executed.append(s)))
+    _rccl.drain(0)
+    assert executed == [0, 1, 2, 3], executed
+    assert rccl.last_arrival() == arrival
+
+    # agreement: all ranks adopt rank 0's arrival order [2, 0, 3, 1]
+    agreed = rccl.scheduler_agree()
+    expect_order = arrivals.get(0) if np >= 1 else list(range(4))
+    assert agreed == expect_order, (agreed, expect_order)
+
+    # round 2: same adversarial arrivals; release follows the agreed order
+    executed2 = []
+    for slot in arrival:
+        _rccl.start_task(0, names[slot],
+                         (lambda s=slot: executed2.append(s)))
+    _rccl.drain(0)
+    assert executed2 == expect_order, executed2
+
+    # cross-rank agreement proof: consensus over the executed sequence
+    blob = bytes(executed2)
+    assert kf.consensus_bytes(blob, "order-check")
+
+    # anonymous tasks bypass ordering but stay on the one dispatcher
+    got = []
+    _rccl.start_task(0, "", lambda: got.append(1))
+    _rccl.drain(0)
+    assert got == [1]
+    rccl.finalize()
+    kf.finalize()
+    return True
+
+
+def rccl_local_scope_agreement_body(rank, np):
+    """LOCAL-scope order agreement uses the intra-host broadcast; on one
+    host (loopback) the local scope covers all ranks."""
+    import kungfu_amd as kf
+    from kungfu_amd import _rccl
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    rccl.init_cpu()
+    assert rccl.scope_size(rccl.LOCAL) == np
+    assert rccl.scope_rank(rccl.LOCAL) == rank
+    # cross scope: one host -> single master (global rank 0)
+    assert rccl.scope_size(rccl.CROSS) == 1
+    assert rccl.scope_member(rccl.CROSS) == (rank == 0)
+    names = ["a", "b"]
+    rccl.scheduler_reset(names, scope=rccl.LOCAL)
+    run = []
+    order = [1, 0] if rank == 0 else [0, 1]
+    for slot in order:
+        _rccl.start_task(rccl.LOCAL, names[slot],
+                         (lambda s=slot: run.append(s)))
+    _rccl.drain(rccl.LOCAL)
+    assert run == [0, 1]
+    agreed = rccl.scheduler_agree(scope=rccl.LOCAL)
+    assert agreed == [1, 0]  # rank 0's arrival order
+    rccl.finalize()
+    kf.finalize()
+    return True
