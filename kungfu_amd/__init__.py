@@ -49,7 +49,11 @@ def init(with_torch=True):
     """Initialize the runtime (idempotent).
 
     Starts the C++ control plane and, when CUDA(HIP) devices and a
-    multi-process cluster are present, the RCCL process group.
+    multi-process cluster are present, the native RCCL collective layer
+    (communicators bootstrapped over the control plane itself — see
+    kungfu_amd.ops.rccl). torch.distributed is only used as an explicit
+    fallback (KUNGFU_GPU_BACKEND=torch) or for gloo-backed CPU tests
+    (KUNGFU_TORCH_BACKEND=gloo).
     """
     global _initialized, _torch_dist
     if _initialized:
@@ -58,11 +62,49 @@ def init(with_torch=True):
     _core.init()
     _initialized = True
     if with_torch:
+        _init_gpu_backend()
+
+
+def _set_cuda_device():
+    import torch
+
+    rank, world = _core.rank(), _core.size()
+    if torch.cuda.device_count() >= world:
+        # launcher-less run (e.g. torchrun on one node)
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+    else:
+        torch.cuda.set_device(0)  # kungfu-run pins HIP_VISIBLE_DEVICES
+
+
+def _init_gpu_backend():
+    """Bring up the GPU collective backend for the current cluster.
+
+    Default: the native RCCL layer (kungfu_amd._rccl) with the uniqueId
+    rendezvous over the C++ control plane — no TCPStore/env:// beside it.
+    KUNGFU_GPU_BACKEND=torch selects the torch.distributed fallback;
+    KUNGFU_TORCH_BACKEND (e.g. gloo) forces a torch process group for
+    CPU-side tests.
+    """
+    if _core.size() <= 1:
+        return
+    import torch
+
+    if os.environ.get("KUNGFU_TORCH_BACKEND"):  # tests: "gloo"
         _maybe_init_torch_dist()
+        return
+    if not torch.cuda.is_available():
+        return  # CPU plumbing mode uses the C++ engine only
+    if os.environ.get("KUNGFU_GPU_BACKEND", "rccl") == "torch":
+        _maybe_init_torch_dist()
+        return
+    _set_cuda_device()
+    from kungfu_amd.ops import rccl as rccl_ops
+
+    rccl_ops.init_gpu()
 
 
 def _maybe_init_torch_dist():
-    """Bring up torch.distributed over RCCL (GPU) for the current cluster."""
+    """Bring up torch.distributed (fallback / gloo test backend)."""
     global _torch_dist
     if _torch_dist or _core.size() <= 1:
         return
@@ -80,11 +122,7 @@ def _maybe_init_torch_dist():
         return
     rank, world = _core.rank(), _core.size()
     if torch.cuda.is_available():
-        if torch.cuda.device_count() >= world:
-            # launcher-less run (e.g. torchrun on one node)
-            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
-        else:
-            torch.cuda.set_device(0)  # kungfu-run pins HIP_VISIBLE_DEVICES
+        _set_cuda_device()
     if os.environ.get("MASTER_ADDR") and os.environ.get("MASTER_PORT"):
         # torchrun path: attach to the launcher's store (env://) — creating
         # our own TCPStore on MASTER_PORT would collide with torchrun's
@@ -133,6 +171,13 @@ def finalize():
     global _initialized, _torch_dist
     if not _initialized:
         return
+    try:
+        # the RCCL layer holds a capsule into _core: tear it down first
+        from kungfu_amd.ops import rccl as rccl_ops
+
+        rccl_ops.finalize()
+    except Exception:
+        pass
     try:
         import torch.distributed as dist
 
@@ -220,7 +265,15 @@ def resize(new_size=None):
     else:
         changed, det = _core.resize(int(new_size))
     if changed and not det:
-        _reinit_torch_dist()
+        from kungfu_amd.ops import rccl as rccl_ops
+
+        if rccl_ops.active():
+            # native path: rebuild communicators in place over the new
+            # cluster (reference ResetNcclHelper) — no process-group
+            # destroy/recreate
+            rccl_ops.reinit()
+        else:
+            _reinit_torch_dist()
     return changed, det
 
 

@@ -32,13 +32,40 @@ def _auto_name(kind, numel):
     return "%s%d#%d" % (kind, numel, next(_name_seq[(kind, numel)]))
 
 
+def _native():
+    """The native RCCL layer module when it is the active GPU backend
+    (default on GPU hosts), else None (torch.distributed fallback)."""
+    from kungfu_amd.ops import rccl as _r
+
+    return _r if _r.active() else None
+
+
+class _NativeWork:
+    """torch.distributed.Work-alike over a native RCCL handle: wait()
+    orders the current stream after the collective (no host block)."""
+
+    __slots__ = ("_handle",)
+
+    def __init__(self, handle):
+        self._handle = handle
+
+    def wait(self):
+        if self._handle is not None:
+            from kungfu_amd.ops import rccl as _r
+
+            _r.wait(self._handle)
+            self._handle = None
+        return True
+
+
 def _dist():
     import torch.distributed as dist
 
     if not dist.is_initialized():
         raise RuntimeError(
-            "CUDA collective requested but the RCCL process group is not "
-            "up; call kungfu_amd.init() first (no silent CPU fallback)")
+            "CUDA collective requested but no GPU backend is up (native "
+            "RCCL inactive, torch.distributed uninitialized); call "
+            "kungfu_amd.init() first (no silent CPU fallback)")
     return dist
 
 
@@ -66,6 +93,17 @@ def all_reduce(tensor, op="sum", name=None, average=False, async_op=False):
     if _core.size() == 1:
         return tensor
     if tensor.is_cuda:
+        nat = _native()
+        if nat is not None:
+            h = nat.all_reduce_async(tensor, op=op, name=name or "")
+            if async_op:
+                if average:
+                    raise ValueError("average not supported with async_op")
+                return tensor, _NativeWork(h)
+            nat.wait(h)
+            if average:
+                tensor.div_(_core.size())
+            return tensor
         dist = _dist()
         work = dist.all_reduce(tensor, op=_torch_reduce_op(op),
                                async_op=async_op)
@@ -90,6 +128,9 @@ def broadcast(tensor, root=0, name=None):
     if _core.size() == 1:
         return tensor
     if tensor.is_cuda:
+        nat = _native()
+        if nat is not None:
+            return nat.broadcast(tensor, root=root, name=name or "")
         _dist().broadcast(tensor, src=root)
         return tensor
     t = tensor.contiguous()
@@ -111,6 +152,12 @@ def all_gather(tensor, name=None):
         out[0] = tensor
         return out
     if tensor.is_cuda:
+        nat = _native()
+        if nat is not None:
+            nat.wait(nat.all_gather_async(out.view(np_, -1),
+                                          tensor.contiguous().view(-1),
+                                          name=name or ""))
+            return out
         dist = _dist()
         dist.all_gather_into_tensor(out.view(np_, -1),
                                     tensor.contiguous().view(1, -1))
@@ -129,6 +176,9 @@ def reduce(tensor, op="sum", name=None):
     if _core.size() == 1:
         return tensor
     if tensor.is_cuda:
+        nat = _native()
+        if nat is not None:
+            return nat.reduce(tensor, op=op, root=0, name=name or "")
         _dist().reduce(tensor, dst=0, op=_torch_reduce_op(op))
         return tensor
     t = tensor.contiguous()
@@ -219,6 +269,12 @@ def hierarchical_all_reduce(tensor, name=None):
     _ensure_init()
     if _core.host_count() <= 1:
         return all_reduce(tensor, name=name)
+    if tensor.is_cuda:
+        nat = _native()
+        if nat is not None:
+            # native sub-communicators (LOCAL + CROSS scopes) — no torch
+            # sub-groups involved
+            return nat.hierarchical_all_reduce(tensor, name=name or "")
     if dist.is_available() and dist.is_initialized():
         g = _hierarchical_groups()
         dist.reduce(tensor, dst=g["local_master"], group=g["local"])
@@ -256,6 +312,14 @@ def broadcast_parameters(params, root=0):
     if not params:
         return
     if params[0].is_cuda:
+        nat = _native()
+        if nat is not None:
+            handles = [nat.broadcast_async(p.data, root=root,
+                                           name="bcast/%d" % i)
+                       for i, p in enumerate(params)]
+            for h in handles:
+                nat.wait(h)
+            return
         dist = _dist()
         for p in params:
             dist.broadcast(p.data, src=root)

@@ -114,8 +114,6 @@ class PairAveragingOptimizer(KungFuOptimizer):
         GPU-to-GPU model sendrecv over RCCL (no host staging), fused
         on-device averaging. Both partners run the same schedule, so the
         exchange is deadlock-free by construction."""
-        import torch.distributed as dist
-
         n, r = _core.size(), _core.rank()
         partner = tournament_partner(r, self._rr_step, n)
         self._rr_step += 1
@@ -126,10 +124,19 @@ class PairAveragingOptimizer(KungFuOptimizer):
         g.pack()  # flat <- v
         if self._recv_buf is None:
             self._recv_buf = torch.empty_like(g.flat)
-        ops = [dist.P2POp(dist.isend, g.flat, partner),
-               dist.P2POp(dist.irecv, self._recv_buf, partner)]
-        for work in dist.batch_isend_irecv(ops):
-            work.wait()
+        from kungfu_amd.ops import _native
+
+        nat = _native()
+        if nat is not None:
+            nat.wait(nat.send_recv_async(g.flat, self._recv_buf, partner,
+                                         name="gossip"))
+        else:
+            import torch.distributed as dist
+
+            ops = [dist.P2POp(dist.isend, g.flat, partner),
+                   dist.P2POp(dist.irecv, self._recv_buf, partner)]
+            for work in dist.batch_isend_irecv(ops):
+                work.wait()
         if g.flat.is_cuda:
             from kungfu_amd.ops import hip as hip_ops
 

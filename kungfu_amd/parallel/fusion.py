@@ -46,6 +46,23 @@ def _alias_view(flat_slice, p):
     return flat_slice.view_as(p)
 
 
+class _NativeBucketWork:
+    """Work-alike over a native RCCL handle (stream-ordered wait)."""
+
+    __slots__ = ("_h",)
+
+    def __init__(self, h):
+        self._h = h
+
+    def wait(self):
+        if self._h is not None:
+            from kungfu_amd.ops import rccl as _r
+
+            _r.wait(self._h)
+            self._h = None
+        return True
+
+
 class _Bucket:
     __slots__ = ("index", "params", "flat", "numel", "ready", "work",
                  "launched", "param_flat", "momentum")
@@ -138,16 +155,29 @@ class GradBucketReducer:
         if b.ready == len(b.params):
             self._drain()
 
+    def _launch_bucket(self, b):
+        """Issue the bucket's async all-reduce on the active GPU backend:
+        native RCCL (default — stream-ordered handle through the per-scope
+        ordered dispatcher) or torch.distributed (fallback)."""
+        from kungfu_amd.ops import _native
+
+        nat = _native()
+        if nat is not None:
+            h = nat.all_reduce_async(b.flat,
+                                     name="%s/%d" % (self.name, b.index))
+            return _NativeBucketWork(h)
+        import torch.distributed as dist
+
+        return dist.all_reduce(b.flat, async_op=True)
+
     def _drain(self):
         # launch complete buckets strictly in bucket-index order so the
         # RCCL op order is identical on every rank
-        import torch.distributed as dist
-
         while self._next_launch < len(self.buckets):
             b = self.buckets[self._next_launch]
             if b.ready < len(b.params):
                 return
-            b.work = dist.all_reduce(b.flat, async_op=True)
+            b.work = self._launch_bucket(b)
             b.launched = True
             self._next_launch += 1
 
@@ -165,19 +195,17 @@ class GradBucketReducer:
         if self.world <= 1:
             return
         if self.is_cuda:
-            import torch.distributed as dist
-
             if self.overlap:
                 self._drain()
                 for b in self.buckets:
                     if not b.launched:  # param got no grad this step
-                        b.work = dist.all_reduce(b.flat, async_op=True)
+                        b.work = self._launch_bucket(b)
                         b.launched = True
                 for b in self.buckets:
                     b.work.wait()
             else:
                 for b in self.buckets:
-                    dist.all_reduce(b.flat)
+                    self._launch_bucket(b).wait()
             if self.average:
                 for b in self.buckets:
                     b.flat.div_(self.world)

@@ -759,3 +759,91 @@ def rccl_local_scope_agreement_body(rank, np):
     rccl.finalize()
     kf.finalize()
     return True
+
+
+def rccl_gpu_world1_body(rank, np):
+    """Full native-RCCL path on one GPU: bootstrap (uniqueId over the
+    control plane), every op, stream-ordered waits, reinit, finalize."""
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    torch.cuda.set_device(0)
+    rccl.init_gpu(0)
+    assert rccl.active()
+    assert rccl.scope_size(rccl.GLOBAL) == np
+    dev = torch.device("cuda:0")
+    x = torch.arange(1024, dtype=torch.float32, device=dev)
+    ref = x.clone()
+    rccl.all_reduce(x)
+    torch.cuda.synchronize()
+    assert torch.equal(x, ref * np)
+    b = torch.full((257,), float(rank), device=dev)
+    rccl.broadcast(b, root=0)
+    torch.cuda.synchronize()
+    assert torch.equal(b, torch.zeros_like(b))
+    g = rccl.all_gather(torch.full((3,), float(rank + 1), device=dev))
+    torch.cuda.synchronize()
+    for r in range(np):
+        assert float(g[r, 0]) == r + 1
+    # bf16 all-reduce (the bucket dtype on the training hot path)
+    xb = torch.ones(4096, dtype=torch.bfloat16, device=dev)
+    rccl.all_reduce(xb)
+    torch.cuda.synchronize()
+    assert float(xb[0]) == float(np)
+    # async handle + host wait
+    h = rccl.all_reduce_async(x, name="async1")
+    rccl.wait_host(h)
+    assert torch.equal(x, ref * np * np)
+    # self/partner sendrecv (gossip primitive)
+    s = torch.full((64,), float(rank * 10 + 7), device=dev)
+    r_ = torch.zeros(64, device=dev)
+    partner = (rank + 1) % np
+    rccl.wait(rccl.send_recv_async(s, r_, partner, name="sr"))
+    torch.cuda.synchronize()
+    expect = float(partner * 10 + 7)
+    assert float(r_[0]) == expect, (float(r_[0]), expect)
+    # hierarchical degenerates to local scopes on one host but must work
+    hx = torch.ones(128, device=dev)
+    rccl.hierarchical_all_reduce(hx)
+    torch.cuda.synchronize()
+    assert float(hx[0]) == float(np)
+    # reinit (elastic path) and use the rebuilt communicator
+    rccl.reinit()
+    y = torch.ones(16, device=dev)
+    rccl.all_reduce(y)
+    torch.cuda.synchronize()
+    assert float(y[0]) == float(np)
+    rccl.finalize()
+    kf.finalize()
+    return True
+
+
+def rccl_gpu_pair_body(rank, np):
+    """Two ranks sharing ONE device: RCCL may refuse duplicate GPUs in a
+    communicator (NCCL semantics). Either a working comm or a clean,
+    agreeing error on both ranks is a pass; a hang is the only failure."""
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    torch.cuda.set_device(0)
+    try:
+        rccl.init_gpu(0)
+    except RuntimeError as e:
+        kf.finalize()
+        return "unsupported: %s" % str(e)[:80]
+    x = torch.ones(1 << 20, device="cuda:0")
+    try:
+        rccl.all_reduce(x)
+        torch.cuda.synchronize()
+        ok = float(x[0]) == float(np)
+    except RuntimeError as e:
+        rccl.finalize()
+        kf.finalize()
+        return "unsupported: %s" % str(e)[:80]
+    rccl.finalize()
+    kf.finalize()
+    return "ok" if ok else "bad-sum"
