@@ -35,9 +35,12 @@ def parse_args():
                    choices=["sync", "sma", "pair", "gns", "sma-gns"])
     p.add_argument("--seq-len", type=int, default=128)
     p.add_argument("--dtype", default="bf16",
-                   choices=["bf16", "bf16-pure", "fp32"],
-                   help="bf16 = fp32 master weights + autocast bf16 compute"
-                        " (fastest measured); bf16-pure = all-bf16 model")
+                   choices=["bf16", "bf16-master", "bf16-pure", "fp32"],
+                   help="bf16 = autocast (fp32 weights, bf16 compute); "
+                        "bf16-master = bf16 conv/linear weights with f32 "
+                        "masters inside the fused optimizer (no per-step "
+                        "cast kernels, half-size all-reduce); "
+                        "bf16-pure = all-bf16 model")
     p.add_argument("--bucket-mb", type=int, default=32)
     p.add_argument("--no-overlap", action="store_true")
     p.add_argument("--graph", type=int, default=None,
@@ -66,12 +69,19 @@ def build_model_and_data(args, device, dtype, amp):
     if args.model == "resnet50":
         from kungfu_amd.models import resnet50
 
-        use_fused = bool(args.fused_bn) and amp and device.type == "cuda"
+        master = args.dtype == "bf16-master"
+        use_fused = bool(args.fused_bn) and (amp or master) and \
+            device.type == "cuda"
         model = resnet50(fused_bn=use_fused)
         x = torch.randn(args.batch_size, 3, 224, 224)
         y = torch.randint(0, 1000, (args.batch_size,), device=device)
         model = model.to(device=device, dtype=dtype)
         x = x.to(device=device, dtype=dtype)
+        if master:
+            from kungfu_amd.utils.precision import convert_bf16_master
+
+            convert_bf16_master(model)
+            x = x.to(torch.bfloat16)
         if args.channels_last and device.type == "cuda":
             model = model.to(memory_format=torch.channels_last)
             x = x.contiguous(memory_format=torch.channels_last)
@@ -112,12 +122,17 @@ def build_model_and_data(args, device, dtype, amp):
     if args.model == "bert":
         from kungfu_amd.models import bert_base
 
-        use_fused_ln = amp and device.type == "cuda"
+        master = args.dtype == "bf16-master"
+        use_fused_ln = (amp or master) and device.type == "cuda"
         model = bert_base(max_len=max(args.seq_len, 128),
                           fused_ln=use_fused_ln).to(
             device=device, dtype=dtype)
         if amp:
             model = model.to(dtype=torch.float32)
+        if master:
+            from kungfu_amd.utils.precision import convert_bf16_master
+
+            convert_bf16_master(model)
         ids = torch.randint(0, 30522, (args.batch_size, args.seq_len),
                             device=device)
         labels = torch.randint(0, 30522,
@@ -220,7 +235,8 @@ def main():
     amp = args.dtype == "bf16"
     dtype = torch.bfloat16 if args.dtype == "bf16-pure" else torch.float32
     if args.channels_last is None:
-        args.channels_last = 1 if amp else 0
+        args.channels_last = 1 if args.dtype in ("bf16",
+                                                 "bf16-master") else 0
 
     torch.manual_seed(1234 + rank)
     model, step_fn, per_gpu_batch = build_model_and_data(args, device,

@@ -285,6 +285,33 @@ __global__ void sgd_momentum_kernel(T *__restrict__ p,
     }
 }
 
+// Mixed-precision variant: model weights (and grads) in T (bf16), the
+// authoritative copy + momentum in f32. One kernel replaces the autocast
+// master-weight pattern's separate cast passes (bf16<->f32 copies every
+// step); traffic is 20 B/elem either way but the per-step cast kernels
+// (~0.5 ms on ResNet-50 b64) disappear and the all-reduce payload halves.
+template <typename T>
+__global__ void sgd_momentum_master_kernel(
+    T *__restrict__ p, const T *__restrict__ g,
+    float *__restrict__ master, float *__restrict__ m, long long n,
+    float lr, float momentum, float weight_decay, float grad_scale,
+    int nesterov)
+{
+    const long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         i < n; i += stride) {
+        const float pi = master[i];
+        const float gi = Elem<T>::to_f(g[i]) * grad_scale +
+                         weight_decay * pi;
+        const float mi = momentum * m[i] + gi;
+        m[i] = mi;
+        const float upd = nesterov ? gi + momentum * mi : mi;
+        const float np = pi - lr * upd;
+        master[i] = np;
+        p[i] = Elem<T>::from_f(np);
+    }
+}
+
 }  // namespace
 
 // ---------------------------------------------------------------------------
@@ -350,6 +377,16 @@ void launch_sgd(void *p, const void *g, float *m, long long n, float lr,
     hipLaunchKernelGGL(sgd_momentum_kernel<T>, dim3(elementwise_grid(n, 8)),
                        dim3(BLOCK), 0, s, (T *)p, (const T *)g, m, n, lr, mu,
                        wd, gs, nesterov);
+}
+template <typename T>
+void launch_sgd_master(void *p, const void *g, float *master, float *m,
+                       long long n, float lr, float mu, float wd, float gs,
+                       int nesterov, hipStream_t s)
+{
+    hipLaunchKernelGGL(sgd_momentum_master_kernel<T>,
+                       dim3(elementwise_grid(n, 8)), dim3(BLOCK), 0, s,
+                       (T *)p, (const T *)g, master, m, n, lr, mu, wd, gs,
+                       nesterov);
 }
 template <typename T>
 void launch_transform2(int op, void *z, const void *x, long long n,
@@ -431,6 +468,18 @@ hipError_t kf_sgd_momentum(void *p, const void *g, void *m_f32, long long n,
 {
     DISPATCH(dtype, launch_sgd, p, g, (float *)m_f32, n, lr, momentum,
              weight_decay, grad_scale, nesterov, (hipStream_t)stream);
+    return hipGetLastError();
+}
+
+hipError_t kf_sgd_momentum_master(void *p, const void *g, void *master_f32,
+                                  void *m_f32, long long n, float lr,
+                                  float momentum, float weight_decay,
+                                  float grad_scale, int nesterov, int dtype,
+                                  void *stream)
+{
+    DISPATCH(dtype, launch_sgd_master, p, g, (float *)master_f32,
+             (float *)m_f32, n, lr, momentum, weight_decay, grad_scale,
+             nesterov, (hipStream_t)stream);
     return hipGetLastError();
 }
 

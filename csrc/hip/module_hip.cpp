@@ -40,6 +40,9 @@ hipError_t kf_dot(const void *, const void *, long long, void *, int,
                   void *);
 hipError_t kf_sgd_momentum(void *, const void *, void *, long long, float,
                            float, float, float, int, int, void *);
+hipError_t kf_sgd_momentum_master(void *, const void *, void *, void *,
+                                  long long, float, float, float, float,
+                                  int, int, void *);
 hipError_t kf_transform2(void *, const void *, long long, int, int, void *);
 hipError_t kf_ln_fwd(const void *, void *, const void *, const void *,
                      void *, void *, long long, int, float, void *);
@@ -187,6 +190,18 @@ PYBIND11_MODULE(_hip, m)
                                     nesterov ? 1 : 0, dtype,
                                     (void *)stream),
                     "kf_sgd_momentum");
+          });
+    m.def("sgd_momentum_master",
+          [](uintptr_t p, uintptr_t g, uintptr_t master_f32,
+             uintptr_t m_f32, long long n, float lr, float momentum,
+             float weight_decay, float grad_scale, bool nesterov, int dtype,
+             uintptr_t stream) {
+              check(kf_sgd_momentum_master(
+                        (void *)p, (const void *)g, (void *)master_f32,
+                        (void *)m_f32, n, lr, momentum, weight_decay,
+                        grad_scale, nesterov ? 1 : 0, dtype,
+                        (void *)stream),
+                    "kf_sgd_momentum_master");
           });
     m.def("transform2",
           [](uintptr_t z, uintptr_t x, long long n, int op, int dtype,

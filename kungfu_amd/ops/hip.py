@@ -174,6 +174,24 @@ def sgd_momentum(param, grad, momentum_buf, lr, momentum=0.9,
                       hip_dtype(param.dtype), _stream())
 
 
+def sgd_momentum_master(param, grad, master, momentum_buf, lr, momentum=0.9,
+                        weight_decay=0.0, grad_scale=1.0, nesterov=False):
+    """Mixed-precision fused SGD: bf16 params/grads, f32 master + momentum.
+    Replaces the autocast master-weight pattern's per-step cast passes."""
+    _require()
+    _check_cuda(param, grad, master, momentum_buf)
+    assert param.numel() == grad.numel() == master.numel() \
+        == momentum_buf.numel()
+    assert master.dtype == torch.float32
+    assert momentum_buf.dtype == torch.float32
+    _hip.sgd_momentum_master(param.data_ptr(), grad.data_ptr(),
+                             master.data_ptr(), momentum_buf.data_ptr(),
+                             param.numel(), float(lr), float(momentum),
+                             float(weight_decay), float(grad_scale),
+                             bool(nesterov), hip_dtype(param.dtype),
+                             _stream())
+
+
 def transform2(z, x, op="sum"):
     """z <- z op x elementwise on device."""
     _require()

@@ -54,6 +54,12 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
                 off = (off + p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
             b.param_flat = flat
             b.momentum = mom
+            # low-precision buckets (bf16-master mode): the authoritative
+            # f32 copy lives here and the fused kernel keeps param/master
+            # in sync — no separate cast passes (vs autocast's per-step
+            # bf16<->f32 master-weight copies)
+            b.master = (flat.to(torch.float32)
+                        if flat.dtype != torch.float32 else None)
 
     def _fused_apply(self):
         from kungfu_amd.ops import hip as hip_ops
@@ -61,11 +67,18 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
         g = self.optimizer.param_groups[0]
         scale = self.reducer.grad_scale
         for b in self.reducer.buckets:
-            hip_ops.sgd_momentum(
-                b.param_flat, b.flat, b.momentum, lr=g["lr"],
-                momentum=g.get("momentum", 0.0),
-                weight_decay=g.get("weight_decay", 0.0), grad_scale=scale,
-                nesterov=g.get("nesterov", False))
+            if b.master is not None:
+                hip_ops.sgd_momentum_master(
+                    b.param_flat, b.flat, b.master, b.momentum, lr=g["lr"],
+                    momentum=g.get("momentum", 0.0),
+                    weight_decay=g.get("weight_decay", 0.0),
+                    grad_scale=scale, nesterov=g.get("nesterov", False))
+            else:
+                hip_ops.sgd_momentum(
+                    b.param_flat, b.flat, b.momentum, lr=g["lr"],
+                    momentum=g.get("momentum", 0.0),
+                    weight_decay=g.get("weight_decay", 0.0),
+                    grad_scale=scale, nesterov=g.get("nesterov", False))
 
     def state_dict(self):
         sd = {"inner": self.optimizer.state_dict()}
@@ -73,6 +86,7 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
             # momentum lives in the fused flat buffers, not in torch state
             sd["fused_momentum"] = [b.momentum for b in
                                     self.reducer.buckets]
+            sd["fused_master"] = [b.master for b in self.reducer.buckets]
         return sd
 
     def load_state_dict(self, sd):
@@ -83,6 +97,10 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
         if self.fused_step and "fused_momentum" in sd:
             for b, m in zip(self.reducer.buckets, sd["fused_momentum"]):
                 b.momentum.copy_(m.to(b.momentum.device))
+        if self.fused_step and "fused_master" in sd:
+            for b, m in zip(self.reducer.buckets, sd["fused_master"]):
+                if b.master is not None and m is not None:
+                    b.master.copy_(m.to(b.master.device))
 
     def _step(self):
         if self.fused_step:

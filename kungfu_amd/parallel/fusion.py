@@ -65,7 +65,7 @@ class _NativeBucketWork:
 
 class _Bucket:
     __slots__ = ("index", "params", "flat", "numel", "ready", "work",
-                 "launched", "param_flat", "momentum")
+                 "launched", "param_flat", "momentum", "master")
 
     def __init__(self, index):
         self.index = index
@@ -75,6 +75,7 @@ class _Bucket:
         self.ready = 0
         self.work = None
         self.launched = False
+        self.master = None
 
 
 class GradBucketReducer:

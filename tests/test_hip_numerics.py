@@ -99,6 +99,30 @@ def test_sgd_momentum(dtype, nesterov):
     assert torch.allclose(m, mf2, atol=1e-5, rtol=1e-5)
 
 
+@pytest.mark.parametrize("nesterov", [False, True])
+def test_sgd_momentum_master(nesterov):
+    """bf16 params/grads with an f32 master: the master must follow the
+    exact f32 update rule and the bf16 param must be its rounding."""
+    n = 100_000
+    torch.manual_seed(2)
+    p = _rand(n, torch.bfloat16)
+    g = _rand(n, torch.bfloat16)
+    master = p.float().clone()
+    m = torch.rand(n, device="cuda", dtype=torch.float32)
+    lr, mu, wd, gs = 0.1, 0.9, 1e-4, 0.5
+    gi = g.float() * gs + wd * master
+    mf2 = mu * m.clone() + gi
+    upd = gi + mu * mf2 if nesterov else mf2
+    ref_master = master - lr * upd
+    hip_ops.sgd_momentum_master(p, g, master, m, lr=lr, momentum=mu,
+                                weight_decay=wd, grad_scale=gs,
+                                nesterov=nesterov)
+    torch.cuda.synchronize()
+    assert torch.allclose(master, ref_master, atol=1e-6, rtol=1e-6)
+    assert torch.equal(p, ref_master.to(torch.bfloat16))
+    assert torch.allclose(m, mf2, atol=1e-5, rtol=1e-5)
+
+
 @pytest.mark.parametrize("op,fn", [
     ("sum", lambda a, b: a + b),
     ("min", torch.minimum),
