@@ -80,17 +80,24 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
     if (row_off < rows_per_blk) {
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-        // 2-row ILP: two independent 16-B loads in flight per iteration
-        for (; r + row_step < M; r += 2 * row_step) {
+        // 4-row ILP: four independent 16-B loads in flight per iteration
+        // (memory-latency hiding needs deep MLP per lane, G13)
+        for (; r + 3 * row_step < M; r += 4 * row_step) {
             const ushort8 va = *(const ushort8 *)(x + r * C + g * 8);
             const ushort8 vb =
                 *(const ushort8 *)(x + (r + row_step) * C + g * 8);
+            const ushort8 vc =
+                *(const ushort8 *)(x + (r + 2 * row_step) * C + g * 8);
+            const ushort8 vd =
+                *(const ushort8 *)(x + (r + 3 * row_step) * C + g * 8);
 #pragma unroll
             for (int k = 0; k < 8; ++k) {
                 const float fa = b2f(va[k]);
                 const float fb = b2f(vb[k]);
-                s[k] += fa + fb;
-                q[k] += fa * fa + fb * fb;
+                const float fc = b2f(vc[k]);
+                const float fd = b2f(vd[k]);
+                s[k] += (fa + fb) + (fc + fd);
+                q[k] += (fa * fa + fb * fb) + (fc * fc + fd * fd);
             }
         }
         for (; r < M; r += row_step) {
@@ -175,8 +182,45 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
         br[k] = b[g * 8 + k];
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
-    for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-         r < M; r += row_step) {
+    long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+    // 2-row ILP: keep 2 (4 with residual) loads in flight per lane
+    for (; r + row_step < M; r += 2 * row_step) {
+        const long long ba = r * C + (long long)g * 8;
+        const long long bb = (r + row_step) * C + (long long)g * 8;
+        const ushort8 va = *(const ushort8 *)(x + ba);
+        const ushort8 vb = *(const ushort8 *)(x + bb);
+        ushort8 rva, rvb;
+        if (RES) {
+            rva = *(const ushort8 *)(res + ba);
+            rvb = *(const ushort8 *)(res + bb);
+        }
+        ushort8 oa, ob;
+        unsigned char ma = 0, mb2 = 0;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            float ua = fmaf(ar[k], b2f(va[k]), br[k]);
+            float ub = fmaf(ar[k], b2f(vb[k]), br[k]);
+            if (RES) {
+                ua += b2f(rva[k]);
+                ub += b2f(rvb[k]);
+            }
+            if (RELU) {
+                if (ua > 0.f) ma |= (unsigned char)(1u << k);
+                if (ub > 0.f) mb2 |= (unsigned char)(1u << k);
+                ua = fmaxf(ua, 0.f);
+                ub = fmaxf(ub, 0.f);
+            }
+            oa[k] = f2b(ua);
+            ob[k] = f2b(ub);
+        }
+        *(ushort8 *)(y + ba) = oa;
+        *(ushort8 *)(y + bb) = ob;
+        if (RELU && mask) {
+            mask[r * gpr + g] = ma;
+            mask[(r + row_step) * gpr + g] = mb2;
+        }
+    }
+    for (; r < M; r += row_step) {
         const long long base = r * C + (long long)g * 8;
         const ushort8 v = *(const ushort8 *)(x + base);
         ushort8 rv;
@@ -232,29 +276,49 @@ __global__ void bn_bwd_reduce_kernel(
         }
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-        // 2-row ILP: four independent 16-B loads in flight per iteration
-        for (; r + row_step < M; r += 2 * row_step) {
+        // 4-row ILP: eight independent 16-B loads in flight per iteration
+        for (; r + 3 * row_step < M; r += 4 * row_step) {
             const long long ba = r * C + (long long)g * 8;
             const long long bb = (r + row_step) * C + (long long)g * 8;
+            const long long bc =
+                (r + 2 * row_step) * C + (long long)g * 8;
+            const long long bd =
+                (r + 3 * row_step) * C + (long long)g * 8;
             const ushort8 dva = *(const ushort8 *)(dy + ba);
             const ushort8 xva = *(const ushort8 *)(x + ba);
             const ushort8 dvb = *(const ushort8 *)(dy + bb);
             const ushort8 xvb = *(const ushort8 *)(x + bb);
+            const ushort8 dvc = *(const ushort8 *)(dy + bc);
+            const ushort8 xvc = *(const ushort8 *)(x + bc);
+            const ushort8 dvd = *(const ushort8 *)(dy + bd);
+            const ushort8 xvd = *(const ushort8 *)(x + bd);
             const unsigned char ma =
                 MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
             const unsigned char mbm =
                 MASKED ? mask[(r + row_step) * gpr + g]
                        : (unsigned char)0xff;
+            const unsigned char mc =
+                MASKED ? mask[(r + 2 * row_step) * gpr + g]
+                       : (unsigned char)0xff;
+            const unsigned char md =
+                MASKED ? mask[(r + 3 * row_step) * gpr + g]
+                       : (unsigned char)0xff;
 #pragma unroll
             for (int k = 0; k < 8; ++k) {
                 float da = b2f(dva[k]);
                 float db = b2f(dvb[k]);
+                float dc = b2f(dvc[k]);
+                float dd = b2f(dvd[k]);
                 if (MASKED && !((ma >> k) & 1)) da = 0.f;
                 if (MASKED && !((mbm >> k) & 1)) db = 0.f;
+                if (MASKED && !((mc >> k) & 1)) dc = 0.f;
+                if (MASKED && !((md >> k) & 1)) dd = 0.f;
                 const float xha = (b2f(xva[k]) - mr[k]) * rr[k];
                 const float xhb = (b2f(xvb[k]) - mr[k]) * rr[k];
-                s1[k] += da + db;
-                s2[k] += da * xha + db * xhb;
+                const float xhc = (b2f(xvc[k]) - mr[k]) * rr[k];
+                const float xhd = (b2f(xvd[k]) - mr[k]) * rr[k];
+                s1[k] += (da + db) + (dc + dd);
+                s2[k] += (da * xha + db * xhb) + (dc * xhc + dd * xhd);
             }
         }
         for (; r < M; r += row_step) {
@@ -324,8 +388,43 @@ __global__ void bn_bwd_dx_kernel(
         t2[k] = sums[C + c] * invM;  // mean of dy_m * xhat
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
-    for (long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-         r < M; r += row_step) {
+    long long r = (long long)blockIdx.x * rows_per_blk + row_off;
+    // 2-row ILP: four 16-B loads in flight per lane
+    for (; r + row_step < M; r += 2 * row_step) {
+        const long long ba = r * C + (long long)g * 8;
+        const long long bb = (r + row_step) * C + (long long)g * 8;
+        const ushort8 dva = *(const ushort8 *)(dy + ba);
+        const ushort8 xva = *(const ushort8 *)(x + ba);
+        const ushort8 dvb = *(const ushort8 *)(dy + bb);
+        const ushort8 xvb = *(const ushort8 *)(x + bb);
+        const unsigned char ma =
+            MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
+        const unsigned char mbm =
+            MASKED ? mask[(r + row_step) * gpr + g] : (unsigned char)0xff;
+        ushort8 dxa, dxb, dra, drb;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+            float da = b2f(dva[k]);
+            float db = b2f(dvb[k]);
+            if (MASKED && !((ma >> k) & 1)) da = 0.f;
+            if (MASKED && !((mbm >> k) & 1)) db = 0.f;
+            if (RES) {
+                dra[k] = f2b(da);
+                drb[k] = f2b(db);
+            }
+            const float xha = (b2f(xva[k]) - mr[k]) * rr[k];
+            const float xhb = (b2f(xvb[k]) - mr[k]) * rr[k];
+            dxa[k] = f2b(ar[k] * (da - t1[k] - xha * t2[k]));
+            dxb[k] = f2b(ar[k] * (db - t1[k] - xhb * t2[k]));
+        }
+        *(ushort8 *)(dx + ba) = dxa;
+        *(ushort8 *)(dx + bb) = dxb;
+        if (RES) {
+            *(ushort8 *)(dres + ba) = dra;
+            *(ushort8 *)(dres + bb) = drb;
+        }
+    }
+    for (; r < M; r += row_step) {
         const long long base = r * C + (long long)g * 8;
         const ushort8 dv = *(const ushort8 *)(dy + base);
         const ushort8 xv = *(const ushort8 *)(x + base);
@@ -362,9 +461,10 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
     long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
-    // >= 64 KiB per block so the atomic epilogue stays cold on small layers
-    const long long by_bytes = (M * C * 2 + 65535) / 65536;
-    if (blocks > by_bytes) blocks = by_bytes;
+    // fill the chip: 2048 blocks = 8 waves-of-4 per CU (the old
+    // 64-KiB-per-block floor starved small-C layers to ~1.5 blocks/CU and
+    // left the kernel latency-bound at ~1.5 TB/s); the NSHADOW-interleaved
+    // atomic epilogue keeps contention cold at any block count
     if (blocks > 2048) blocks = 2048;
     if (blocks < 1) blocks = 1;
     hipLaunchKernelGGL(bn_stats_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
@@ -422,8 +522,6 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
     long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
-    const long long by_bytes = (M * C * 4 + 65535) / 65536;  // 2 streams
-    if (blocks > by_bytes) blocks = by_bytes;
     if (blocks > 2048) blocks = 2048;
     if (blocks < 1) blocks = 1;
     const dim3 grid((uint32_t)blocks), block(BLOCK);

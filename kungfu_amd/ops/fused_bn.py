@@ -31,18 +31,21 @@ def _nhwc_ok(x):
 class _FusedBNFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, residual, weight, bias, running_mean, running_var,
-                momentum, eps, relu, training):
+                momentum, eps, relu, training, ws):
         C = x.shape[1]
         M = x.numel() // C
         s = _stream()
         dev = x.device
         if training:
-            sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
+            # persistent per-module workspace: zero_() only, no per-call
+            # allocations (was ~100 torch.zeros launches/step on ResNet-50)
+            sums = ws["fwd_sums"]
+            sums.zero_()
             _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)
-            save_mean = torch.empty(C, dtype=torch.float32, device=dev)
-            save_rstd = torch.empty(C, dtype=torch.float32, device=dev)
-            a = torch.empty(C, dtype=torch.float32, device=dev)
-            b = torch.empty(C, dtype=torch.float32, device=dev)
+            save_mean = ws["save_mean"]
+            save_rstd = ws["save_rstd"]
+            a = ws["a"]
+            b = ws["b"]
             _hip.bn_finalize(sums.data_ptr(), weight.data_ptr(),
                              bias.data_ptr(), running_mean.data_ptr(),
                              running_var.data_ptr(), save_mean.data_ptr(),
@@ -70,6 +73,7 @@ class _FusedBNFunction(torch.autograd.Function):
         ctx.has_res = residual is not None
         ctx.has_mask = mask is not None
         ctx.MC = (M, C)
+        ctx.ws = ws
         return y
 
     @staticmethod
@@ -80,8 +84,8 @@ class _FusedBNFunction(torch.autograd.Function):
         if dy.dtype != x.dtype:  # kernels require bf16 dy (matching x)
             dy = dy.to(x.dtype)
         dy = dy.contiguous(memory_format=torch.channels_last)
-        dev = x.device
-        sums = torch.zeros(16 * C, dtype=torch.float32, device=dev)
+        sums = ctx.ws["bwd_sums"]
+        sums.zero_()
         mask_ptr = mask.data_ptr() if ctx.has_mask else 0
         _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), mask_ptr,
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
@@ -95,7 +99,8 @@ class _FusedBNFunction(torch.autograd.Function):
                        dres.data_ptr() if dres is not None else 0, s)
         db = sums[:C]               # db = sum(dy_m)
         dw = sums[C:2 * C]          # dw = sum(dy_m * xhat)
-        return (dx, dres, dw, db, None, None, None, None, None, None)
+        return (dx, dres, dw, db, None, None, None, None, None, None,
+                None)
 
 
 class FusedBNReLU2d(torch.nn.Module):
@@ -112,6 +117,25 @@ class FusedBNReLU2d(torch.nn.Module):
         self.bias = torch.nn.Parameter(torch.zeros(channels))
         self.register_buffer("running_mean", torch.zeros(channels))
         self.register_buffer("running_var", torch.ones(channels))
+        self._ws = None
+
+    def _workspace(self, dev):
+        """Persistent kernel workspace (shadow accumulators, folded
+        scale/shift, saved stats): fixed addresses keep hipGraph capture
+        stable and remove ~4 small allocations+fills per call. Relies on
+        the standard one-forward-one-backward pairing per step."""
+        if self._ws is None or self._ws["a"].device != dev:
+            C = self.channels
+            f32 = torch.float32
+            self._ws = {
+                "fwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
+                "bwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
+                "save_mean": torch.empty(C, dtype=f32, device=dev),
+                "save_rstd": torch.empty(C, dtype=f32, device=dev),
+                "a": torch.empty(C, dtype=f32, device=dev),
+                "b": torch.empty(C, dtype=f32, device=dev),
+            }
+        return self._ws
 
     def _params_f32(self):
         # the kernels reinterpret these pointers as float32: after e.g.
@@ -129,7 +153,7 @@ class FusedBNReLU2d(torch.nn.Module):
             return _FusedBNFunction.apply(
                 x, residual, self.weight, self.bias, self.running_mean,
                 self.running_var, self.momentum, self.eps, self.relu,
-                self.training)
+                self.training, self._workspace(x.device))
         # eager fallback (CPU, odd shapes): numerically the reference
         y = torch.nn.functional.batch_norm(
             x.float(), self.running_mean, self.running_var, self.weight,

@@ -100,23 +100,35 @@ class GradBucketReducer:
     # -- construction ---------------------------------------------------
 
     def _build_buckets(self, bucket_bytes):
-        # reverse parameter order: grads become ready roughly back-to-front
+        # reverse parameter order: grads become ready roughly back-to-front.
+        # Buckets are dtype-homogeneous; mixed-precision models (bf16-master
+        # mode: bf16 conv/linear weights interleaved with f32 norm params)
+        # keep ONE OPEN BUCKET PER DTYPE so alternating dtypes don't
+        # fragment into dozens of tiny buckets (collective-per-bucket).
         order = list(reversed(self.params))
         self.buckets = []
-        cur = _Bucket(0)
+        open_by_dtype = {}
+        # launch order = bucket CLOSE order: a bucket that fills early in
+        # backward launches early; the per-dtype buckets still open at the
+        # end close last. Derived from the static param list alone, so it
+        # is identical on every rank by construction (the determinism the
+        # reference's NCCLScheduler arrival-order broadcast provides).
+        self._launch_seq = []
         for p in order:
-            # capacity in elements of THIS bucket's dtype (buckets are
-            # homogeneous; a dtype change forces a bucket break)
             cap = max(1, bucket_bytes // p.element_size())
-            dtype_break = cur.params and cur.params[0].dtype != p.dtype
             aligned = (p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
-            if cur.numel > 0 and (cur.numel + aligned > cap or dtype_break):
-                self.buckets.append(cur)
+            cur = open_by_dtype.get(p.dtype)
+            if cur is None or cur.numel + aligned > cap:
+                if cur is not None:
+                    self._launch_seq.append(cur.index)  # closed: launch slot
                 cur = _Bucket(len(self.buckets))
+                self.buckets.append(cur)
+                open_by_dtype[p.dtype] = cur
             cur.params.append(p)
             cur.numel += aligned
-        if cur.params:
-            self.buckets.append(cur)
+        for b in self.buckets:
+            if b.index not in self._launch_seq:
+                self._launch_seq.append(b.index)
         # allocate flats and alias grads
         self.bucket_of = {}
         for b in self.buckets:
@@ -172,10 +184,10 @@ class GradBucketReducer:
         return dist.all_reduce(b.flat, async_op=True)
 
     def _drain(self):
-        # launch complete buckets strictly in bucket-index order so the
-        # RCCL op order is identical on every rank
-        while self._next_launch < len(self.buckets):
-            b = self.buckets[self._next_launch]
+        # launch complete buckets strictly in the precomputed launch
+        # sequence so the RCCL op order is identical on every rank
+        while self._next_launch < len(self._launch_seq):
+            b = self.buckets[self._launch_seq[self._next_launch]]
             if b.ready < len(b.params):
                 return
             b.work = self._launch_bucket(b)
@@ -198,15 +210,16 @@ class GradBucketReducer:
         if self.is_cuda:
             if self.overlap:
                 self._drain()
-                for b in self.buckets:
+                for i in self._launch_seq:
+                    b = self.buckets[i]
                     if not b.launched:  # param got no grad this step
                         b.work = self._launch_bucket(b)
                         b.launched = True
                 for b in self.buckets:
                     b.work.wait()
             else:
-                for b in self.buckets:
-                    self._launch_bucket(b).wait()
+                for i in self._launch_seq:
+                    self._launch_bucket(self.buckets[i]).wait()
             if self.average:
                 for b in self.buckets:
                     b.flat.div_(self.world)
