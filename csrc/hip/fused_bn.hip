@@ -25,9 +25,22 @@
 namespace {
 
 constexpr int BLOCK = 256;
-// Cross-block reduction uses NSHADOW interleaved accumulator copies to cut
-// same-address atomic contention by 8x; bn_finalize / bn_fold sum them.
-constexpr int NSHADOW = 8;
+// Cross-block reduction is ATOMIC-FREE: every block writes its 2C partial
+// sums to partials[blockIdx][2C] (plain stores) and a parallel fold kernel
+// reduces the partial rows. Global atomics at full-chip block counts
+// (2048 blocks x 2C adds) measured 2x slower on the big early layers; the
+// partial buffer is bounded to ~2 MiB by capping blocks at 262144/C.
+constexpr int MAX_PARTIAL_FLOATS = 2 * 262144;  // 2 MiB scratch
+
+inline long long bn_reduce_blocks(long long M, int C, int rows_per_blk)
+{
+    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
+    const long long cap = 262144 / C;  // partials <= 2 MiB
+    if (blocks > cap) blocks = cap;
+    if (blocks > 2048) blocks = 2048;
+    if (blocks < 1) blocks = 1;
+    return blocks;
+}
 
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
 typedef __attribute__((ext_vector_type(4))) float float4v;
@@ -116,10 +129,33 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
         atomicAdd(&lds[C + g * 8 + k], q[k]);
     }
     __syncthreads();
-    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
-    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-        atomicAdd(&shadow[i], lds[i]);
+    // atomic-free epilogue: plain store of this block's partial row
+    float *row = sums + (size_t)blockIdx.x * 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) row[i] = lds[i];
+}
+
+// Fold partial rows: out[c] = sum over nblocks of partials[r][c].
+// 16 channels x 16 row-lanes per block; LDS tree over the lanes.
+__global__ void bn_reduce_partials_kernel(
+    const float *__restrict__ partials, int nblocks, int C2,
+    float *__restrict__ out)
+{
+    constexpr int CPB = 16;
+    const int c = blockIdx.x * CPB + ((int)threadIdx.x % CPB);
+    const int rl = (int)threadIdx.x / CPB;  // 0..15
+    __shared__ float lds[BLOCK];
+    float acc = 0.f;
+    if (c < C2) {
+        for (int r = rl; r < nblocks; r += BLOCK / CPB)
+            acc += partials[(size_t)r * C2 + c];
     }
+    lds[threadIdx.x] = acc;
+    __syncthreads();
+    for (int s = (BLOCK / CPB) / 2; s > 0; s >>= 1) {
+        if (rl < s) lds[threadIdx.x] += lds[threadIdx.x + s * CPB];
+        __syncthreads();
+    }
+    if (rl == 0 && c < C2) out[c] = lds[threadIdx.x];
 }
 
 // ---- finalize: mean/var -> folded scale/shift + running stats ----
@@ -138,11 +174,8 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
 {
     const int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
-    float s0 = 0.f, s1 = 0.f;
-    for (int k = 0; k < NSHADOW; ++k) {
-        s0 += sums[(size_t)k * 2 * C + c];
-        s1 += sums[(size_t)k * 2 * C + C + c];
-    }
+    const float s0 = sums[c];      // folded by bn_reduce_partials
+    const float s1 = sums[C + c];
     const float mean = s0 / (float)M;
     const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
     const float rstd = rsqrtf(var + eps);
@@ -344,20 +377,8 @@ __global__ void bn_bwd_reduce_kernel(
         atomicAdd(&lds[C + g * 8 + k], s2[k]);
     }
     __syncthreads();
-    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
-    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-        atomicAdd(&shadow[i], lds[i]);
-    }
-}
-
-// Fold the NSHADOW accumulator copies into copy 0.
-__global__ void bn_fold_kernel(float *__restrict__ sums, int C)
-{
-    const int i = blockIdx.x * blockDim.x + threadIdx.x;
-    if (i >= 2 * C) return;
-    float acc = sums[i];
-    for (int k = 1; k < NSHADOW; ++k) acc += sums[(size_t)k * 2 * C + i];
-    sums[i] = acc;
+    float *row = sums + (size_t)blockIdx.x * 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) row[i] = lds[i];
 }
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
@@ -454,22 +475,20 @@ __global__ void bn_bwd_dx_kernel(
 
 extern "C" {
 
-hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
-                       void *stream)
+hipError_t kf_bn_stats(const void *x, long long M, int C, void *partials,
+                       void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
-    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
-    // fill the chip: 2048 blocks = 8 waves-of-4 per CU (the old
-    // 64-KiB-per-block floor starved small-C layers to ~1.5 blocks/CU and
-    // left the kernel latency-bound at ~1.5 TB/s); the NSHADOW-interleaved
-    // atomic epilogue keeps contention cold at any block count
-    if (blocks > 2048) blocks = 2048;
-    if (blocks < 1) blocks = 1;
+    const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
     hipLaunchKernelGGL(bn_stats_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
                        2 * C * sizeof(float), (hipStream_t)stream,
-                       (const unsigned short *)x, M, C, (float *)sums);
+                       (const unsigned short *)x, M, C, (float *)partials);
+    hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                       dim3((2 * C + 15) / 16), dim3(BLOCK), 0,
+                       (hipStream_t)stream, (const float *)partials,
+                       (int)blocks, 2 * C, (float *)sums);
     return hipGetLastError();
 }
 
@@ -516,14 +535,12 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
 hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                             const void *mask, const void *mean,
                             const void *rstd, long long M, int C,
-                            void *sums, void *stream)
+                            void *partials, void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
-    long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
-    if (blocks > 2048) blocks = 2048;
-    if (blocks < 1) blocks = 1;
+    const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
     const size_t lds = 2 * C * sizeof(float);
@@ -532,17 +549,14 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                        (const unsigned short *)dy,                          \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)mean,    \
-                       (const float *)rstd, M, C, (float *)sums)
+                       (const float *)rstd, M, C, (float *)partials)
     if (mask) CASE(true);
     else CASE(false);
 #undef CASE
-    return hipGetLastError();
-}
-
-hipError_t kf_bn_fold(void *sums, int C, void *stream)
-{
-    hipLaunchKernelGGL(bn_fold_kernel, dim3((2 * C + 255) / 256), dim3(256),
-                       0, (hipStream_t)stream, (float *)sums, C);
+    hipLaunchKernelGGL(bn_reduce_partials_kernel,
+                       dim3((2 * C + 15) / 16), dim3(BLOCK), 0, s,
+                       (const float *)partials, (int)blocks, 2 * C,
+                       (float *)sums);
     return hipGetLastError();
 }
 
