@@ -15,35 +15,53 @@
 //
 // Per-channel reductions: each lane owns one channel-octet in registers
 // (f32x8 accumulators), reduces into LDS with shared-memory atomics, one
-// global atomic per channel per block (Guideline 12). With M = N*H*W in
-// the tens of thousands and <= 2048 blocks, global atomics are cold.
+// global atomic per channel per block into NSHADOW interleaved shadow
+// copies (Guideline 12). Measured alternatives that LOST: full-chip grids
+// with the same atomics (2x slower on the big layers — atomic storms),
+// and atomic-free per-block partial rows + a fold kernel (fold is
+// latency-bound at ~21 us/call). Geometry stays env-tunable for sweeps:
+//   KF_BN_CAPKB  — min KiB of input per block (0 = off); default 64
+//   KF_BN_MAXBLK — hard block cap; default 2048
+//   KF_BN_ILP    — rows in flight per lane in reduction kernels (2|4)
 #include <hip/hip_bf16.h>
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 namespace {
 
 constexpr int BLOCK = 256;
-// Cross-block reduction is ATOMIC-FREE: every block writes its 2C partial
-// sums to partials[blockIdx][2C] (plain stores) and a parallel fold kernel
-// reduces the partial rows. Global atomics at full-chip block counts
-// (2048 blocks x 2C adds) measured 2x slower on the big early layers; the
-// partial buffer is bounded to ~2 MiB by capping blocks at 262144/C.
-constexpr int MAX_PARTIAL_FLOATS = 2 * 262144;  // 2 MiB scratch
+constexpr int NSHADOW = 8;
+
+inline int env_int(const char *name, int dflt)
+{
+    const char *v = getenv(name);
+    return v && *v ? atoi(v) : dflt;
+}
 
 inline long long bn_reduce_blocks(long long M, int C, int rows_per_blk)
 {
+    static const int cap_kb = env_int("KF_BN_CAPKB", 64);
+    static const int maxblk = env_int("KF_BN_MAXBLK", 2048);
     long long blocks = (M + rows_per_blk - 1) / rows_per_blk;
-    const long long cap = 262144 / C;  // partials <= 2 MiB
-    if (blocks > cap) blocks = cap;
-    if (blocks > 2048) blocks = 2048;
+    if (cap_kb > 0) {
+        const long long by_bytes =
+            (M * C * 2 + cap_kb * 1024 - 1) / (cap_kb * 1024);
+        if (blocks > by_bytes) blocks = by_bytes;
+    }
+    if (blocks > maxblk) blocks = maxblk;
     if (blocks < 1) blocks = 1;
     return blocks;
 }
 
+inline int bn_ilp()
+{
+    static const int ilp = env_int("KF_BN_ILP", 2);
+    return ilp >= 4 ? 4 : 2;
+}
+
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
-typedef __attribute__((ext_vector_type(4))) float float4v;
 
 __device__ inline float b2f(unsigned short u)
 {
@@ -66,16 +84,9 @@ __device__ inline unsigned short f2b(float f)
     return (unsigned short)(r >> 16);
 }
 
-inline int grid_for(long long work_items)
-{
-    long long g = (work_items + BLOCK - 1) / BLOCK;
-    if (g < 1) g = 1;
-    if (g > 2048) g = 2048;
-    return (int)g;
-}
-
 // ---- pass 1: per-channel sum / sumsq ----
-// x: [M][C] bf16; out sums: f32[2*C] {sum, sumsq} (pre-zeroed)
+// x: [M][C] bf16; out sums: f32[NSHADOW][2*C] (pre-zeroed)
+template <int ILP>
 __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
                                 long long M, int C,
                                 float *__restrict__ sums)
@@ -93,24 +104,21 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
     if (row_off < rows_per_blk) {
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-        // 4-row ILP: four independent 16-B loads in flight per iteration
-        // (memory-latency hiding needs deep MLP per lane, G13)
-        for (; r + 3 * row_step < M; r += 4 * row_step) {
-            const ushort8 va = *(const ushort8 *)(x + r * C + g * 8);
-            const ushort8 vb =
-                *(const ushort8 *)(x + (r + row_step) * C + g * 8);
-            const ushort8 vc =
-                *(const ushort8 *)(x + (r + 2 * row_step) * C + g * 8);
-            const ushort8 vd =
-                *(const ushort8 *)(x + (r + 3 * row_step) * C + g * 8);
+        // ILP rows in flight per lane (independent 16-B loads)
+        for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
+            ushort8 v[ILP];
 #pragma unroll
-            for (int k = 0; k < 8; ++k) {
-                const float fa = b2f(va[k]);
-                const float fb = b2f(vb[k]);
-                const float fc = b2f(vc[k]);
-                const float fd = b2f(vd[k]);
-                s[k] += (fa + fb) + (fc + fd);
-                q[k] += (fa * fa + fb * fb) + (fc * fc + fd * fd);
+            for (int j = 0; j < ILP; ++j)
+                v[j] = *(const ushort8 *)(x + (r + j * row_step) * C +
+                                          g * 8);
+#pragma unroll
+            for (int j = 0; j < ILP; ++j) {
+#pragma unroll
+                for (int k = 0; k < 8; ++k) {
+                    const float f = b2f(v[j][k]);
+                    s[k] += f;
+                    q[k] += f * f;
+                }
             }
         }
         for (; r < M; r += row_step) {
@@ -129,38 +137,15 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
         atomicAdd(&lds[C + g * 8 + k], q[k]);
     }
     __syncthreads();
-    // atomic-free epilogue: plain store of this block's partial row
-    float *row = sums + (size_t)blockIdx.x * 2 * C;
-    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) row[i] = lds[i];
-}
-
-// Fold partial rows: out[c] = sum over nblocks of partials[r][c].
-// 16 channels x 16 row-lanes per block; LDS tree over the lanes.
-__global__ void bn_reduce_partials_kernel(
-    const float *__restrict__ partials, int nblocks, int C2,
-    float *__restrict__ out)
-{
-    constexpr int CPB = 16;
-    const int c = blockIdx.x * CPB + ((int)threadIdx.x % CPB);
-    const int rl = (int)threadIdx.x / CPB;  // 0..15
-    __shared__ float lds[BLOCK];
-    float acc = 0.f;
-    if (c < C2) {
-        for (int r = rl; r < nblocks; r += BLOCK / CPB)
-            acc += partials[(size_t)r * C2 + c];
+    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+        atomicAdd(&shadow[i], lds[i]);
     }
-    lds[threadIdx.x] = acc;
-    __syncthreads();
-    for (int s = (BLOCK / CPB) / 2; s > 0; s >>= 1) {
-        if (rl < s) lds[threadIdx.x] += lds[threadIdx.x + s * CPB];
-        __syncthreads();
-    }
-    if (rl == 0 && c < C2) out[c] = lds[threadIdx.x];
 }
 
 // ---- finalize: mean/var -> folded scale/shift + running stats ----
-// sums: {sum,sumsq}; outputs a = w*rstd, b = bias - mean*a; saves
-// mean/rstd for backward; updates running stats in place (momentum).
+// sums: NSHADOW x {sum,sumsq}; outputs a = w*rstd, b = bias - mean*a;
+// saves mean/rstd for backward; updates running stats (momentum).
 __global__ void bn_finalize_kernel(const float *__restrict__ sums,
                                    const float *__restrict__ weight,
                                    const float *__restrict__ bias,
@@ -174,8 +159,11 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
 {
     const int c = blockIdx.x * blockDim.x + threadIdx.x;
     if (c >= C) return;
-    const float s0 = sums[c];      // folded by bn_reduce_partials
-    const float s1 = sums[C + c];
+    float s0 = 0.f, s1 = 0.f;
+    for (int k = 0; k < NSHADOW; ++k) {
+        s0 += sums[(size_t)k * 2 * C + c];
+        s1 += sums[(size_t)k * 2 * C + C + c];
+    }
     const float mean = s0 / (float)M;
     const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
     const float rstd = rsqrtf(var + eps);
@@ -194,7 +182,7 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
 
 // ---- pass 2: y = [relu]( a*x + b [+ res] ) ----
 // Fixed channel-octet per thread: per-channel params live in registers for
-// the whole row loop; each block streams contiguous 2*BLOCK*8-byte spans.
+// the whole row loop; each block streams contiguous spans. 2-row ILP.
 template <bool RELU, bool RES>
 __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
                               const unsigned short *__restrict__ res,
@@ -216,7 +204,6 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
     long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-    // 2-row ILP: keep 2 (4 with residual) loads in flight per lane
     for (; r + row_step < M; r += 2 * row_step) {
         const long long ba = r * C + (long long)g * 8;
         const long long bb = (r + row_step) * C + (long long)g * 8;
@@ -278,10 +265,10 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
 }
 
 // ---- backward pass 1: per-channel sums of dy_m and dy_m * xhat ----
-// dy_m = dy * relu_mask (mask recomputed from x [+res]); xhat from
-// save_mean/save_rstd. out: f32[2*C] {sum_dy, sum_dyxhat} (pre-zeroed:
-// these ARE db and dw).
-template <bool MASKED>
+// dy_m = dy * relu_mask; xhat from save_mean/save_rstd.
+// out: f32[NSHADOW][2*C] {sum_dy, sum_dyxhat} (pre-zeroed; folded copies
+// ARE db and dw).
+template <int ILP, bool MASKED>
 __global__ void bn_bwd_reduce_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
@@ -309,49 +296,28 @@ __global__ void bn_bwd_reduce_kernel(
         }
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-        // 4-row ILP: eight independent 16-B loads in flight per iteration
-        for (; r + 3 * row_step < M; r += 4 * row_step) {
-            const long long ba = r * C + (long long)g * 8;
-            const long long bb = (r + row_step) * C + (long long)g * 8;
-            const long long bc =
-                (r + 2 * row_step) * C + (long long)g * 8;
-            const long long bd =
-                (r + 3 * row_step) * C + (long long)g * 8;
-            const ushort8 dva = *(const ushort8 *)(dy + ba);
-            const ushort8 xva = *(const ushort8 *)(x + ba);
-            const ushort8 dvb = *(const ushort8 *)(dy + bb);
-            const ushort8 xvb = *(const ushort8 *)(x + bb);
-            const ushort8 dvc = *(const ushort8 *)(dy + bc);
-            const ushort8 xvc = *(const ushort8 *)(x + bc);
-            const ushort8 dvd = *(const ushort8 *)(dy + bd);
-            const ushort8 xvd = *(const ushort8 *)(x + bd);
-            const unsigned char ma =
-                MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
-            const unsigned char mbm =
-                MASKED ? mask[(r + row_step) * gpr + g]
-                       : (unsigned char)0xff;
-            const unsigned char mc =
-                MASKED ? mask[(r + 2 * row_step) * gpr + g]
-                       : (unsigned char)0xff;
-            const unsigned char md =
-                MASKED ? mask[(r + 3 * row_step) * gpr + g]
-                       : (unsigned char)0xff;
+        for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
+            ushort8 dv[ILP], xv[ILP];
+            unsigned char mk[ILP];
 #pragma unroll
-            for (int k = 0; k < 8; ++k) {
-                float da = b2f(dva[k]);
-                float db = b2f(dvb[k]);
-                float dc = b2f(dvc[k]);
-                float dd = b2f(dvd[k]);
-                if (MASKED && !((ma >> k) & 1)) da = 0.f;
-                if (MASKED && !((mbm >> k) & 1)) db = 0.f;
-                if (MASKED && !((mc >> k) & 1)) dc = 0.f;
-                if (MASKED && !((md >> k) & 1)) dd = 0.f;
-                const float xha = (b2f(xva[k]) - mr[k]) * rr[k];
-                const float xhb = (b2f(xvb[k]) - mr[k]) * rr[k];
-                const float xhc = (b2f(xvc[k]) - mr[k]) * rr[k];
-                const float xhd = (b2f(xvd[k]) - mr[k]) * rr[k];
-                s1[k] += (da + db) + (dc + dd);
-                s2[k] += (da * xha + db * xhb) + (dc * xhc + dd * xhd);
+            for (int j = 0; j < ILP; ++j) {
+                const long long base =
+                    (r + j * row_step) * C + (long long)g * 8;
+                dv[j] = *(const ushort8 *)(dy + base);
+                xv[j] = *(const ushort8 *)(x + base);
+                mk[j] = MASKED ? mask[(r + j * row_step) * gpr + g]
+                               : (unsigned char)0xff;
+            }
+#pragma unroll
+            for (int j = 0; j < ILP; ++j) {
+#pragma unroll
+                for (int k = 0; k < 8; ++k) {
+                    float d = b2f(dv[j][k]);
+                    if (MASKED && !((mk[j] >> k) & 1)) d = 0.f;
+                    const float xh = (b2f(xv[j][k]) - mr[k]) * rr[k];
+                    s1[k] += d;
+                    s2[k] += d * xh;
+                }
             }
         }
         for (; r < M; r += row_step) {
@@ -377,12 +343,25 @@ __global__ void bn_bwd_reduce_kernel(
         atomicAdd(&lds[C + g * 8 + k], s2[k]);
     }
     __syncthreads();
-    float *row = sums + (size_t)blockIdx.x * 2 * C;
-    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) row[i] = lds[i];
+    float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
+    for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
+        atomicAdd(&shadow[i], lds[i]);
+    }
+}
+
+// Fold the NSHADOW accumulator copies into copy 0.
+__global__ void bn_fold_kernel(float *__restrict__ sums, int C)
+{
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= 2 * C) return;
+    float acc = sums[i];
+    for (int k = 1; k < NSHADOW; ++k) acc += sums[(size_t)k * 2 * C + i];
+    sums[i] = acc;
 }
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
 // dx = w*rstd * (dy_m - sum_dy/M - xhat * sum_dyxhat/M); d_res = dy_m.
+// 2-row ILP.
 template <bool MASKED, bool RES>
 __global__ void bn_bwd_dx_kernel(
     const unsigned short *__restrict__ dy,
@@ -410,7 +389,6 @@ __global__ void bn_bwd_dx_kernel(
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
     long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-    // 2-row ILP: four 16-B loads in flight per lane
     for (; r + row_step < M; r += 2 * row_step) {
         const long long ba = r * C + (long long)g * 8;
         const long long bb = (r + row_step) * C + (long long)g * 8;
@@ -475,20 +453,23 @@ __global__ void bn_bwd_dx_kernel(
 
 extern "C" {
 
-hipError_t kf_bn_stats(const void *x, long long M, int C, void *partials,
-                       void *sums, void *stream)
+hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
+                       void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int gpr = C / 8;
     const int rows_per_blk = BLOCK / gpr;
     const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
-    hipLaunchKernelGGL(bn_stats_kernel, dim3((uint32_t)blocks), dim3(BLOCK),
-                       2 * C * sizeof(float), (hipStream_t)stream,
-                       (const unsigned short *)x, M, C, (float *)partials);
-    hipLaunchKernelGGL(bn_reduce_partials_kernel,
-                       dim3((2 * C + 15) / 16), dim3(BLOCK), 0,
-                       (hipStream_t)stream, (const float *)partials,
-                       (int)blocks, 2 * C, (float *)sums);
+    const dim3 grid((uint32_t)blocks), block(BLOCK);
+    const size_t lds = 2 * C * sizeof(float);
+    const auto s = (hipStream_t)stream;
+    if (bn_ilp() == 4) {
+        hipLaunchKernelGGL((bn_stats_kernel<4>), grid, block, lds, s,
+                           (const unsigned short *)x, M, C, (float *)sums);
+    } else {
+        hipLaunchKernelGGL((bn_stats_kernel<2>), grid, block, lds, s,
+                           (const unsigned short *)x, M, C, (float *)sums);
+    }
     return hipGetLastError();
 }
 
@@ -535,7 +516,7 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
 hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                             const void *mask, const void *mean,
                             const void *rstd, long long M, int C,
-                            void *partials, void *sums, void *stream)
+                            void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int gpr = C / 8;
@@ -544,19 +525,27 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
     const size_t lds = 2 * C * sizeof(float);
-#define CASE(MK)                                                            \
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<MK>), grid, block, lds, s,     \
+#define CASE(I, MK)                                                         \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<I, MK>), grid, block, lds, s,  \
                        (const unsigned short *)dy,                          \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)mean,    \
-                       (const float *)rstd, M, C, (float *)partials)
-    if (mask) CASE(true);
-    else CASE(false);
+                       (const float *)rstd, M, C, (float *)sums)
+    if (bn_ilp() == 4) {
+        if (mask) CASE(4, true);
+        else CASE(4, false);
+    } else {
+        if (mask) CASE(2, true);
+        else CASE(2, false);
+    }
 #undef CASE
-    hipLaunchKernelGGL(bn_reduce_partials_kernel,
-                       dim3((2 * C + 15) / 16), dim3(BLOCK), 0, s,
-                       (const float *)partials, (int)blocks, 2 * C,
-                       (float *)sums);
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_fold(void *sums, int C, void *stream)
+{
+    hipLaunchKernelGGL(bn_fold_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (float *)sums, C);
     return hipGetLastError();
 }
 

@@ -17,8 +17,7 @@ namespace py = pybind11;
 
 extern "C" {
 hipError_t kf_pack(const void *, int, void *, int, void *);
-hipError_t kf_bn_stats(const void *, long long, int, void *, void *,
-                       void *);
+hipError_t kf_bn_stats(const void *, long long, int, void *, void *);
 hipError_t kf_bn_finalize(const void *, const void *, const void *, void *,
                           void *, void *, void *, void *, void *, long long,
                           int, float, float, void *);
@@ -26,7 +25,8 @@ hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
                      const void *, long long, int, int, void *, void *);
 hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, long long, int,
-                            void *, void *, void *);
+                            void *, void *);
+hipError_t kf_bn_fold(void *, int, void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
                         const void *, long long, int, void *, void *,
@@ -238,10 +238,10 @@ PYBIND11_MODULE(_hip, m)
 
     // ---- fused BatchNorm(+residual+ReLU), NHWC bf16 ----
     m.def("bn_stats",
-          [](uintptr_t x, long long M, int C, uintptr_t partials,
-             uintptr_t sums, uintptr_t stream) {
-              check(kf_bn_stats((const void *)x, M, C, (void *)partials,
-                                (void *)sums, (void *)stream),
+          [](uintptr_t x, long long M, int C, uintptr_t sums,
+             uintptr_t stream) {
+              check(kf_bn_stats((const void *)x, M, C, (void *)sums,
+                                (void *)stream),
                     "kf_bn_stats");
           });
     m.def("bn_finalize",
@@ -268,16 +268,18 @@ PYBIND11_MODULE(_hip, m)
           });
     m.def("bn_bwd_reduce",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t mean,
-             uintptr_t rstd, long long M, int C, uintptr_t partials,
-             uintptr_t sums, uintptr_t stream) {
+             uintptr_t rstd, long long M, int C, uintptr_t sums,
+             uintptr_t stream) {
               check(kf_bn_bwd_reduce((const void *)dy, (const void *)x,
                                      (const void *)mask,
                                      (const void *)mean,
                                      (const void *)rstd, M, C,
-                                     (void *)partials, (void *)sums,
-                                     (void *)stream),
+                                     (void *)sums, (void *)stream),
                     "kf_bn_bwd_reduce");
           });
+    m.def("bn_fold", [](uintptr_t sums, int C, uintptr_t stream) {
+        check(kf_bn_fold((void *)sums, C, (void *)stream), "kf_bn_fold");
+    });
     m.def("bn_bwd_dx",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t a,
              uintptr_t mean, uintptr_t rstd, uintptr_t sums, long long M,
