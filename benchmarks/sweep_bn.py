@@ -57,13 +57,15 @@ print(json.dumps(out))
 '''
 
 CONFIGS = [
-    {"KF_BN_CAPKB": "64", "KF_BN_ILP": "2"},   # round-1 baseline
-    {"KF_BN_CAPKB": "64", "KF_BN_ILP": "4"},
-    {"KF_BN_CAPKB": "16", "KF_BN_ILP": "2"},
-    {"KF_BN_CAPKB": "16", "KF_BN_ILP": "4"},
-    {"KF_BN_CAPKB": "256", "KF_BN_ILP": "2"},
-    {"KF_BN_CAPKB": "0", "KF_BN_MAXBLK": "1024", "KF_BN_ILP": "2"},
-    {"KF_BN_CAPKB": "0", "KF_BN_MAXBLK": "512", "KF_BN_ILP": "4"},
+    {"KF_BN_ILP_STATS": "2", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "1"},  # r1
+    {"KF_BN_ILP_STATS": "4", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "1"},
+    {"KF_BN_ILP_STATS": "2", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "2"},
+    {"KF_BN_ILP_STATS": "4", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "2"},
+    {"KF_BN_ILP_STATS": "4", "KF_BN_ILP_BWD": "4", "KF_BN_OCT": "2"},
+    {"KF_BN_ILP_STATS": "4", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "2",
+     "KF_BN_CAPKB": "128"},
+    {"KF_BN_ILP_STATS": "4", "KF_BN_ILP_BWD": "2", "KF_BN_OCT": "2",
+     "KF_BN_CAPKB": "32"},
 ]
 
 

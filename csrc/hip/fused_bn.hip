@@ -55,10 +55,26 @@ inline long long bn_reduce_blocks(long long M, int C, int rows_per_blk)
     return blocks;
 }
 
-inline int bn_ilp()
+inline int bn_ilp_stats()
 {
-    static const int ilp = env_int("KF_BN_ILP", 2);
+    static const int ilp = env_int("KF_BN_ILP_STATS",
+                                   env_int("KF_BN_ILP", 4));
     return ilp >= 4 ? 4 : 2;
+}
+
+inline int bn_ilp_bwd()
+{
+    static const int ilp = env_int("KF_BN_ILP_BWD",
+                                   env_int("KF_BN_ILP", 2));
+    return ilp >= 4 ? 4 : 2;
+}
+
+// channel-octets (8 ch / 16 B) per lane: 2 doubles the per-lane loads in
+// flight (32 B) at the cost of 2x accumulator VGPRs. Requires C % 16 == 0.
+inline int bn_oct()
+{
+    static const int oct = env_int("KF_BN_OCT", 1);
+    return oct >= 2 ? 2 : 1;
 }
 
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
@@ -86,55 +102,66 @@ __device__ inline unsigned short f2b(float f)
 
 // ---- pass 1: per-channel sum / sumsq ----
 // x: [M][C] bf16; out sums: f32[NSHADOW][2*C] (pre-zeroed)
-template <int ILP>
+// OCT octets (8 channels / 16 B each) per lane; ILP rows in flight.
+template <int ILP, int OCT>
 __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
                                 long long M, int C,
                                 float *__restrict__ sums)
 {
     extern __shared__ float lds[];  // 2*C floats
-    const int gpr = C / 8;          // channel-octet groups per row
-    const int rows_per_blk = BLOCK / gpr;  // >= 1 given C <= 2048
+    const int gpr = C / (8 * OCT);  // channel groups per row
+    const int rows_per_blk = BLOCK / gpr;
     const int g = threadIdx.x % gpr;
     const int row_off = threadIdx.x / gpr;
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) lds[i] = 0.f;
     __syncthreads();
 
-    float s[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float q[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    float s[8 * OCT], q[8 * OCT];
+#pragma unroll
+    for (int k = 0; k < 8 * OCT; ++k) s[k] = q[k] = 0.f;
     if (row_off < rows_per_blk) {
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-        // ILP rows in flight per lane (independent 16-B loads)
         for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
-            ushort8 v[ILP];
-#pragma unroll
-            for (int j = 0; j < ILP; ++j)
-                v[j] = *(const ushort8 *)(x + (r + j * row_step) * C +
-                                          g * 8);
+            ushort8 v[ILP][OCT];
 #pragma unroll
             for (int j = 0; j < ILP; ++j) {
 #pragma unroll
-                for (int k = 0; k < 8; ++k) {
-                    const float f = b2f(v[j][k]);
-                    s[k] += f;
-                    q[k] += f * f;
+                for (int o = 0; o < OCT; ++o)
+                    v[j][o] = *(const ushort8 *)(
+                        x + (r + j * row_step) * C + g * 8 * OCT + o * 8);
+            }
+#pragma unroll
+            for (int j = 0; j < ILP; ++j) {
+#pragma unroll
+                for (int o = 0; o < OCT; ++o) {
+#pragma unroll
+                    for (int k = 0; k < 8; ++k) {
+                        const float f = b2f(v[j][o][k]);
+                        s[o * 8 + k] += f;
+                        q[o * 8 + k] += f * f;
+                    }
                 }
             }
         }
         for (; r < M; r += row_step) {
-            const ushort8 v = *(const ushort8 *)(x + r * C + g * 8);
 #pragma unroll
-            for (int k = 0; k < 8; ++k) {
-                const float f = b2f(v[k]);
-                s[k] += f;
-                q[k] += f * f;
+            for (int o = 0; o < OCT; ++o) {
+                const ushort8 v = *(const ushort8 *)(x + r * C +
+                                                     g * 8 * OCT + o * 8);
+#pragma unroll
+                for (int k = 0; k < 8; ++k) {
+                    const float f = b2f(v[k]);
+                    s[o * 8 + k] += f;
+                    q[o * 8 + k] += f * f;
+                }
             }
         }
     }
 #pragma unroll
-    for (int k = 0; k < 8; ++k) {
-        atomicAdd(&lds[g * 8 + k], s[k]);
-        atomicAdd(&lds[C + g * 8 + k], q[k]);
+    for (int k = 0; k < 8 * OCT; ++k) {
+        atomicAdd(&lds[g * 8 * OCT + k], s[k]);
+        atomicAdd(&lds[C + g * 8 * OCT + k], q[k]);
     }
     __syncthreads();
     float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
@@ -268,7 +295,7 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
 // dy_m = dy * relu_mask; xhat from save_mean/save_rstd.
 // out: f32[NSHADOW][2*C] {sum_dy, sum_dyxhat} (pre-zeroed; folded copies
 // ARE db and dw).
-template <int ILP, bool MASKED>
+template <int ILP, int OCT, bool MASKED>
 __global__ void bn_bwd_reduce_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
@@ -277,70 +304,90 @@ __global__ void bn_bwd_reduce_kernel(
     long long M, int C, float *__restrict__ sums)
 {
     extern __shared__ float lds[];  // 2*C floats
-    const int gpr = C / 8;
+    const int gpr = C / (8 * OCT);
     const int rows_per_blk = BLOCK / gpr;
     const int g = threadIdx.x % gpr;
     const int row_off = threadIdx.x / gpr;
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) lds[i] = 0.f;
     __syncthreads();
 
-    float s1[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    float s2[8] = {0, 0, 0, 0, 0, 0, 0, 0};
-    if (row_off < rows_per_blk) {
-        float mr[8], rr[8];
+    float s1[8 * OCT], s2[8 * OCT];
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-            const int c = g * 8 + k;
+    for (int k = 0; k < 8 * OCT; ++k) s1[k] = s2[k] = 0.f;
+    if (row_off < rows_per_blk) {
+        float mr[8 * OCT], rr[8 * OCT];
+#pragma unroll
+        for (int k = 0; k < 8 * OCT; ++k) {
+            const int c = g * 8 * OCT + k;
             mr[k] = mean[c];
             rr[k] = rstd[c];
         }
         const long long row_step = (long long)gridDim.x * rows_per_blk;
         long long r = (long long)blockIdx.x * rows_per_blk + row_off;
         for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
-            ushort8 dv[ILP], xv[ILP];
-            unsigned char mk[ILP];
+            ushort8 dv[ILP][OCT], xv[ILP][OCT];
+            unsigned short mk[ILP];
 #pragma unroll
             for (int j = 0; j < ILP; ++j) {
                 const long long base =
-                    (r + j * row_step) * C + (long long)g * 8;
-                dv[j] = *(const ushort8 *)(dy + base);
-                xv[j] = *(const ushort8 *)(x + base);
-                mk[j] = MASKED ? mask[(r + j * row_step) * gpr + g]
-                               : (unsigned char)0xff;
+                    (r + j * row_step) * C + (long long)g * 8 * OCT;
+#pragma unroll
+                for (int o = 0; o < OCT; ++o) {
+                    dv[j][o] = *(const ushort8 *)(dy + base + o * 8);
+                    xv[j][o] = *(const ushort8 *)(x + base + o * 8);
+                }
+                if (MASKED) {
+                    const unsigned char *mrow =
+                        mask + (r + j * row_step) * (gpr * OCT) + g * OCT;
+                    mk[j] = mrow[0];
+                    if (OCT == 2) mk[j] |= (unsigned short)mrow[1] << 8;
+                } else {
+                    mk[j] = 0xffff;
+                }
             }
 #pragma unroll
             for (int j = 0; j < ILP; ++j) {
 #pragma unroll
-                for (int k = 0; k < 8; ++k) {
-                    float d = b2f(dv[j][k]);
-                    if (MASKED && !((mk[j] >> k) & 1)) d = 0.f;
-                    const float xh = (b2f(xv[j][k]) - mr[k]) * rr[k];
-                    s1[k] += d;
-                    s2[k] += d * xh;
+                for (int o = 0; o < OCT; ++o) {
+#pragma unroll
+                    for (int k = 0; k < 8; ++k) {
+                        float d = b2f(dv[j][o][k]);
+                        if (MASKED && !((mk[j] >> (o * 8 + k)) & 1))
+                            d = 0.f;
+                        const float xh =
+                            (b2f(xv[j][o][k]) - mr[o * 8 + k]) *
+                            rr[o * 8 + k];
+                        s1[o * 8 + k] += d;
+                        s2[o * 8 + k] += d * xh;
+                    }
                 }
             }
         }
         for (; r < M; r += row_step) {
-            const long long base = r * C + (long long)g * 8;
-            const ushort8 dv = *(const ushort8 *)(dy + base);
-            const ushort8 xv = *(const ushort8 *)(x + base);
-            const unsigned char mb =
-                MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
+            const long long base = r * C + (long long)g * 8 * OCT;
 #pragma unroll
-            for (int k = 0; k < 8; ++k) {
-                const float xf = b2f(xv[k]);
-                float d = b2f(dv[k]);
-                if (MASKED && !((mb >> k) & 1)) d = 0.f;
-                const float xh = (xf - mr[k]) * rr[k];
-                s1[k] += d;
-                s2[k] += d * xh;
+            for (int o = 0; o < OCT; ++o) {
+                const ushort8 dv = *(const ushort8 *)(dy + base + o * 8);
+                const ushort8 xv = *(const ushort8 *)(x + base + o * 8);
+                const unsigned char mb =
+                    MASKED ? mask[r * (gpr * OCT) + g * OCT + o]
+                           : (unsigned char)0xff;
+#pragma unroll
+                for (int k = 0; k < 8; ++k) {
+                    const float xf = b2f(xv[k]);
+                    float d = b2f(dv[k]);
+                    if (MASKED && !((mb >> k) & 1)) d = 0.f;
+                    const float xh = (xf - mr[o * 8 + k]) * rr[o * 8 + k];
+                    s1[o * 8 + k] += d;
+                    s2[o * 8 + k] += d * xh;
+                }
             }
         }
     }
 #pragma unroll
-    for (int k = 0; k < 8; ++k) {
-        atomicAdd(&lds[g * 8 + k], s1[k]);
-        atomicAdd(&lds[C + g * 8 + k], s2[k]);
+    for (int k = 0; k < 8 * OCT; ++k) {
+        atomicAdd(&lds[g * 8 * OCT + k], s1[k]);
+        atomicAdd(&lds[C + g * 8 * OCT + k], s2[k]);
     }
     __syncthreads();
     float *shadow = sums + (size_t)(blockIdx.x % NSHADOW) * 2 * C;
@@ -457,19 +504,24 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
                        void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
-    const int gpr = C / 8;
+    const int oct = (C % 16 == 0) ? bn_oct() : 1;
+    const int gpr = C / (8 * oct);
     const int rows_per_blk = BLOCK / gpr;
     const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const size_t lds = 2 * C * sizeof(float);
     const auto s = (hipStream_t)stream;
-    if (bn_ilp() == 4) {
-        hipLaunchKernelGGL((bn_stats_kernel<4>), grid, block, lds, s,
-                           (const unsigned short *)x, M, C, (float *)sums);
+#define CASE(I, O)                                                          \
+    hipLaunchKernelGGL((bn_stats_kernel<I, O>), grid, block, lds, s,        \
+                       (const unsigned short *)x, M, C, (float *)sums)
+    if (bn_ilp_stats() == 4) {
+        if (oct == 2) CASE(4, 2);
+        else CASE(4, 1);
     } else {
-        hipLaunchKernelGGL((bn_stats_kernel<2>), grid, block, lds, s,
-                           (const unsigned short *)x, M, C, (float *)sums);
+        if (oct == 2) CASE(2, 2);
+        else CASE(2, 1);
     }
+#undef CASE
     return hipGetLastError();
 }
 
@@ -519,24 +571,25 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                             void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
-    const int gpr = C / 8;
+    const int oct = (C % 16 == 0) ? bn_oct() : 1;
+    const int gpr = C / (8 * oct);
     const int rows_per_blk = BLOCK / gpr;
     const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
     const size_t lds = 2 * C * sizeof(float);
-#define CASE(I, MK)                                                         \
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<I, MK>), grid, block, lds, s,  \
-                       (const unsigned short *)dy,                          \
+#define CASE(I, O, MK)                                                      \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<I, O, MK>), grid, block, lds, \
+                       s, (const unsigned short *)dy,                       \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)mean,    \
                        (const float *)rstd, M, C, (float *)sums)
-    if (bn_ilp() == 4) {
-        if (mask) CASE(4, true);
-        else CASE(4, false);
+    if (bn_ilp_bwd() == 4) {
+        if (oct == 2) { if (mask) CASE(4, 2, true); else CASE(4, 2, false); }
+        else { if (mask) CASE(4, 1, true); else CASE(4, 1, false); }
     } else {
-        if (mask) CASE(2, true);
-        else CASE(2, false);
+        if (oct == 2) { if (mask) CASE(2, 2, true); else CASE(2, 2, false); }
+        else { if (mask) CASE(2, 1, true); else CASE(2, 1, false); }
     }
 #undef CASE
     return hipGetLastError();
