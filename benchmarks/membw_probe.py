@@ -4,9 +4,14 @@ Times torch's own kernels (copy, sum, add) plus our bn_stats/bn_bwd_reduce
 on the ResNet stem shape, so kernel efficiency is judged against what the
 machine actually delivers, not the paper peak.
 """
+import os
+import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 
 def t(fn, iters=20, warm=3):
@@ -30,13 +35,17 @@ def main():
     yf = torch.empty_like(xf)
     gb = n * 2 / 1e9
 
-    r = {}
-    r["copy_bf16 (R+W)"] = (t(lambda: y.copy_(x)), 2 * gb)
-    r["copy_f32 (R+W)"] = (t(lambda: yf.copy_(xf)), 2 * gb)
-    r["sum_bf16 (R)"] = (t(lambda: x.sum()), gb)
-    r["sum_f32 (R)"] = (t(lambda: xf.sum()), gb)
-    r["add_bf16 (2R+W)"] = (t(lambda: torch.add(x, y, out=y)), 3 * gb)
-    r["sum_2d_ch (R)"] = (t(lambda: x.view(M, C).sum(0)), gb)
+    def show(k, sec, bytes_gb):
+        print("%-22s %8.1f us  %6.2f TB/s" % (k, sec * 1e6,
+                                              bytes_gb / sec / 1e3),
+              flush=True)
+
+    show("copy_bf16 (R+W)", t(lambda: y.copy_(x)), 2 * gb)
+    show("copy_f32 (R+W)", t(lambda: yf.copy_(xf)), 2 * gb)
+    show("sum_bf16 (R)", t(lambda: x.sum()), gb)
+    show("sum_f32 (R)", t(lambda: xf.sum()), gb)
+    show("add_bf16 (2R+W)", t(lambda: torch.add(x, y, out=y)), 3 * gb)
+    show("sum_2d_ch (R)", t(lambda: x.view(M, C).sum(0)), gb)
 
     from kungfu_amd import _hip
 
@@ -46,18 +55,15 @@ def main():
     rstd = torch.ones(C, dtype=torch.float32, device="cuda")
     mask = torch.full((M * (C // 8),), 255, dtype=torch.uint8,
                       device="cuda")
-    r["bn_stats (R)"] = (
-        t(lambda: _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)),
-        gb)
+    show("bn_stats (R)",
+         t(lambda: _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)),
+         gb)
     dy = y
-    r["bn_bwd_reduce (2R)"] = (
-        t(lambda: _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(),
-                                     mask.data_ptr(), mean.data_ptr(),
-                                     rstd.data_ptr(), M, C,
-                                     sums.data_ptr(), s)), 2 * gb)
-    for k, (sec, bytes_gb) in r.items():
-        print("%-22s %8.1f us  %6.2f TB/s" % (k, sec * 1e6,
-                                              bytes_gb / sec / 1e3))
+    show("bn_bwd_reduce (2R)",
+         t(lambda: _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(),
+                                      mask.data_ptr(), mean.data_ptr(),
+                                      rstd.data_ptr(), M, C,
+                                      sums.data_ptr(), s)), 2 * gb)
 
 
 if __name__ == "__main__":
