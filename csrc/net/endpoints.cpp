@@ -23,6 +23,80 @@ std::shared_ptr<CollectiveEndpoint::Slot> CollectiveEndpoint::slot(
     return s;
 }
 
+bool CollectiveEndpoint::on_header(const PeerID &src,
+                                   const FrameHeader &h, Conn &conn)
+{
+    if (h.flags & msgflag::ShmRef) {
+        // colocated shm data path: body is a /dev/shm file reference;
+        // read the file straight into the registered destination when one
+        // is waiting
+        std::string path(h.len, '\0');
+        if (!conn.read_body(path.data(), h.len))
+            throw std::runtime_error("short shm-ref read on " + h.name);
+        FILE *fp = std::fopen(path.c_str(), "rb");
+        if (!fp) throw std::runtime_error("shm chunk missing: " + path);
+        std::fseek(fp, 0, SEEK_END);
+        const size_t n = (size_t)std::ftell(fp);
+        std::fseek(fp, 0, SEEK_SET);
+        auto s = slot(src, h.name);
+        std::unique_lock<std::mutex> lk(s->mu);
+        if (s->dst && !s->filled && !s->filling && n == s->dst_len) {
+            s->filling = true;
+            lk.unlock();
+            const size_t got = std::fread(s->dst, 1, n, fp);
+            std::fclose(fp);
+            ::unlink(path.c_str());
+            lk.lock();
+            s->filling = false;
+            if (got != n) {
+                s->dead = true;
+                s->cv.notify_all();
+                throw std::runtime_error("short shm chunk read: " + path);
+            }
+            s->filled = true;
+            s->cv.notify_all();
+            return true;
+        }
+        auto buf = pool_.get(n);
+        lk.unlock();
+        const size_t got = std::fread(buf.data(), 1, n, fp);
+        std::fclose(fp);
+        ::unlink(path.c_str());
+        if (got != n)
+            throw std::runtime_error("short shm chunk read: " + path);
+        lk.lock();
+        s->q.push_back(std::move(buf));
+        s->cv.notify_all();
+        return true;
+    }
+    auto s = slot(src, h.name);
+    std::unique_lock<std::mutex> lk(s->mu);
+    if (s->dst && !s->filled && !s->filling && h.len == s->dst_len) {
+        // zero-copy: socket read lands in the destination buffer
+        s->filling = true;
+        lk.unlock();
+        const bool ok = conn.read_body(s->dst, h.len);
+        lk.lock();
+        s->filling = false;
+        if (!ok) {
+            s->dead = true;
+            s->cv.notify_all();
+            throw std::runtime_error("short body read on " + h.name);
+        }
+        s->filled = true;
+        s->cv.notify_all();
+        return true;
+    }
+    auto buf = pool_.get(h.len);
+    lk.unlock();
+    if (!conn.read_body(buf.data(), h.len))
+        throw std::runtime_error("short body read on " + h.name);
+    lk.lock();
+    s->q.push_back(std::move(buf));
+    s->cv.notify_all();
+    return true;
+}
+
 void CollectiveEndpoint::on_frame(const PeerID &src, Frame &f)
 {
     if (f.flags & msgflag::ShmRef) {
@@ -70,6 +144,7 @@ void CollectiveEndpoint::recv_into(const PeerID &src, const std::string &name,
         if (buf.size() != len)
             throw std::runtime_error("collective size mismatch on " + name);
         std::memcpy(dst, buf.data(), len);
+        pool_.put(std::move(buf));
         return;
     }
     if (s->dead) throw std::runtime_error("endpoint shut down");
@@ -81,15 +156,21 @@ void CollectiveEndpoint::recv_into(const PeerID &src, const std::string &name,
         s->dst = nullptr;
         return;
     }
-    s->dst = nullptr;
     if (!s->q.empty()) {
+        s->dst = nullptr;
         auto buf = std::move(s->q.front());
         s->q.pop_front();
         if (buf.size() != len)
             throw std::runtime_error("collective size mismatch on " + name);
         std::memcpy(dst, buf.data(), len);
+        pool_.put(std::move(buf));
         return;
     }
+    // dead: an in-flight zero-copy body read may still target dst — wait
+    // it out before the caller frees the buffer
+    s->cv.wait(lk, [&] { return !s->filling; });
+    s->dst = nullptr;
+    if (s->filled) return;
     throw std::runtime_error("recv_into aborted (endpoint shut down) on " +
                              name);
 }

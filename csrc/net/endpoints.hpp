@@ -21,6 +21,13 @@ namespace kf {
 class CollectiveEndpoint {
   public:
     void on_frame(const PeerID &src, Frame &f);
+    // Zero-copy receive: consume the frame body straight off the socket —
+    // into the pre-registered destination when one is waiting (no copy at
+    // all), else into a pooled buffer. Always consumes; throws on socket
+    // failure (the server then drops the connection).
+    bool on_header(const PeerID &src, const FrameHeader &h, Conn &conn);
+    // return a consumed buffer to the receive pool
+    void recycle(std::vector<uint8_t> &&v) { pool_.put(std::move(v)); }
     // Copy (or zero-copy if pre-registered before arrival) one message from
     // (src, name) into dst. Throws on size mismatch or shutdown.
     void recv_into(const PeerID &src, const std::string &name, void *dst,
@@ -38,12 +45,14 @@ class CollectiveEndpoint {
         uint8_t *dst = nullptr;              // registered destination
         size_t dst_len = 0;
         bool filled = false;
+        bool filling = false;  // socket read into dst in progress
         bool dead = false;
     };
     std::shared_ptr<Slot> slot(const PeerID &src, const std::string &name);
 
     std::mutex mu_;
     std::unordered_map<std::string, std::shared_ptr<Slot>> slots_;
+    BufPool pool_;
     bool dead_ = false;
 };
 

@@ -91,6 +91,59 @@ bool Conn::send_frame(const std::string &name, uint32_t flags,
     return true;
 }
 
+std::vector<uint8_t> BufPool::get(size_t n)
+{
+    size_t cls = 64;
+    while (cls < n) cls <<= 1;
+    {
+        std::lock_guard<std::mutex> lk(mu_);
+        auto it = classes_.find(cls);
+        if (it != classes_.end() && !it->second.empty()) {
+            auto v = std::move(it->second.back());
+            it->second.pop_back();
+            v.resize(n);
+            return v;
+        }
+    }
+    std::vector<uint8_t> v;
+    v.reserve(cls);
+    v.resize(n);
+    return v;
+}
+
+void BufPool::put(std::vector<uint8_t> &&v)
+{
+    if (v.capacity() < 64) return;
+    size_t cls = 64;
+    while (cls < v.capacity()) cls <<= 1;
+    if (cls != v.capacity()) cls >>= 1;  // conservative: class it fits
+    std::lock_guard<std::mutex> lk(mu_);
+    auto &vec = classes_[cls];
+    if (vec.size() < kMaxPerClass) vec.emplace_back(std::move(v));
+}
+
+bool Conn::read_header(FrameHeader &h)
+{
+    int fd = fd_.load();
+    if (fd < 0) return false;
+    uint32_t name_len;
+    if (!read_all(fd, &name_len, 4)) return false;
+    if (name_len > (1u << 16)) return false;  // sanity
+    h.name.resize(name_len);
+    if (name_len && !read_all(fd, h.name.data(), name_len)) return false;
+    if (!read_all(fd, &h.flags, 4)) return false;
+    if (!read_all(fd, &h.len, 8)) return false;
+    if (h.len > (1ull << 33)) return false;  // 8 GiB sanity cap
+    return true;
+}
+
+bool Conn::read_body(void *dst, size_t len)
+{
+    int fd = fd_.load();
+    if (fd < 0) return false;
+    return len == 0 || read_all(fd, dst, len);
+}
+
 bool Conn::read_frame(Frame &f)
 {
     int fd = fd_.load();
@@ -135,10 +188,13 @@ Server::Server(const PeerID &self, bool use_unix)
 
 Server::~Server() { stop(); }
 
-void Server::start(FrameHandler handler, std::function<bool(uint32_t)> token_ok)
+void Server::start(FrameHandler handler,
+                   std::function<bool(uint32_t)> token_ok,
+                   HeaderHandler header_handler)
 {
     handler_ = std::move(handler);
     token_ok_ = std::move(token_ok);
+    header_handler_ = std::move(header_handler);
 
     // TCP listener on 0.0.0.0:port
     tcp_fd_ = ::socket(AF_INET, SOCK_STREAM, 0);
@@ -243,8 +299,31 @@ void Server::handle_conn(int fd)
         std::lock_guard<std::mutex> lk(mu_);
         conns_.push_back(conn);
     }
-    Frame f;
-    while (!stopping_.load() && conn->read_frame(f)) {
+    FrameHeader h;
+    while (!stopping_.load() && conn->read_header(h)) {
+        {
+            std::lock_guard<std::mutex> lk(mu_);
+            ingress_[hs.src.key()] += h.len + h.name.size() + 16;
+        }
+        // zero-copy fast path: the endpoint may consume the body straight
+        // off the socket into a pre-registered destination
+        if (header_handler_) {
+            bool consumed = false;
+            try {
+                consumed = header_handler_(hs, h, *conn);
+            } catch (const std::exception &e) {
+                std::fprintf(stderr,
+                             "[kungfu] header handler error on '%s': %s\n",
+                             h.name.c_str(), e.what());
+                break;  // socket state unknown: drop the conn
+            }
+            if (consumed) continue;
+        }
+        Frame f;
+        f.name = std::move(h.name);
+        f.flags = h.flags;
+        f.data.resize(h.len);
+        if (!conn->read_body(f.data.data(), f.data.size())) break;
         if (hs.type == ConnType::Ping) {
             conn->send_frame(f.name, f.flags | msgflag::IsResponse,
                             f.data.data(), f.data.size());
@@ -464,6 +543,12 @@ void Client::reset(const std::vector<PeerID> &keeps, uint32_t token)
             ++it;
         }
     }
+}
+
+std::map<uint64_t, uint64_t> Server::ingress_all() const
+{
+    std::lock_guard<std::mutex> lk(mu_);
+    return ingress_;
 }
 
 uint64_t Client::egress_bytes(const PeerID &remote) const

@@ -42,6 +42,25 @@ struct Frame {
     std::vector<uint8_t> data;
 };
 
+struct FrameHeader {
+    std::string name;
+    uint32_t flags = 0;
+    uint64_t len = 0;
+};
+
+// Size-classed reusable buffer pool (reference byte_slice_pool.go): the
+// receive path would otherwise allocate a fresh heap buffer per frame.
+class BufPool {
+  public:
+    std::vector<uint8_t> get(size_t n);
+    void put(std::vector<uint8_t> &&v);
+
+  private:
+    static constexpr size_t kMaxPerClass = 16;
+    std::mutex mu_;
+    std::map<size_t, std::vector<std::vector<uint8_t>>> classes_;
+};
+
 // Thread-safe framed socket wrapper.
 class Conn {
   public:
@@ -52,6 +71,11 @@ class Conn {
     bool send_frame(const std::string &name, uint32_t flags,
                     const void *data, size_t len);
     bool read_frame(Frame &f);  // blocking; false on EOF/error
+    // split read: header first, then the body into caller-owned memory
+    // (zero-copy receive: the socket read lands in the registered
+    // destination buffer, reference handler/collective.go:34-41)
+    bool read_header(FrameHeader &h);
+    bool read_body(void *dst, size_t len);
     void close_fd();
     int fd() const { return fd_; }
     std::mutex &write_mutex() { return wmu_; }
@@ -69,6 +93,11 @@ struct Handshake {
 
 using FrameHandler =
     std::function<void(const Handshake &, Frame &, Conn &)>;
+// Optional header-level handler: may consume the body straight off the
+// socket (zero-copy); returns true when it did. Falling through (false)
+// makes the server read the body into a Frame and call the FrameHandler.
+using HeaderHandler = std::function<bool(const Handshake &,
+                                         const FrameHeader &, Conn &)>;
 
 // Listens on TCP (self.port) and, when enabled, on a Unix socket keyed by
 // port. One handler thread per accepted connection.
@@ -78,9 +107,13 @@ class Server {
     ~Server();
     // token_ok(peer_token) decides whether to accept (elastic fencing)
     void start(FrameHandler handler,
-               std::function<bool(uint32_t)> token_ok);
+               std::function<bool(uint32_t)> token_ok,
+               HeaderHandler header_handler = nullptr);
     void stop();
     static std::string unix_sock_path(const PeerID &peer);
+    // ingress byte accounting per source peer (reference
+    // monitor/counters.go meters both directions)
+    std::map<uint64_t, uint64_t> ingress_all() const;
 
   private:
     void accept_loop(int listen_fd);
@@ -91,10 +124,12 @@ class Server {
     int tcp_fd_ = -1, unix_fd_ = -1;
     std::atomic<bool> stopping_{false};
     FrameHandler handler_;
+    HeaderHandler header_handler_;
     std::function<bool(uint32_t)> token_ok_;
     std::vector<std::thread> threads_;
-    std::mutex mu_;
+    mutable std::mutex mu_;
     std::vector<std::shared_ptr<Conn>> conns_;
+    std::map<uint64_t, uint64_t> ingress_;
 };
 
 // Outgoing connection pool: one cached conn per (remote, type).

@@ -189,6 +189,12 @@ void Peer::start()
                 // fencing: accept current or newer tokens; reject stale
                 // peers from before the last resize (connection.go:81-87)
                 return token + 1 >= version_;
+            },
+            [this](const Handshake &hs, const FrameHeader &h, Conn &conn) {
+                // zero-copy fast path: collective payloads land straight
+                // in the registered destination (or a pooled buffer)
+                if (hs.type != ConnType::Collective) return false;
+                return collective_.on_header(hs.src, h, conn);
             });
     }
     int r = workers_.rank_of(cfg_.self);
@@ -207,6 +213,14 @@ void Peer::start()
                     p.port = (uint16_t)(kv.first & 0xffff);
                     out += "kungfu_egress_bytes_total{peer=\"" + p.str() +
                            "\"} " + std::to_string(kv.second) + "\n";
+                }
+                for (auto &kv : ingress_bytes()) {
+                    PeerID p;
+                    p.ipv4 = (uint32_t)(kv.first >> 16);
+                    p.port = (uint16_t)(kv.first & 0xffff);
+                    out += "kungfu_ingress_bytes_total{peer=\"" +
+                           p.str() + "\"} " + std::to_string(kv.second) +
+                           "\n";
                 }
                 return out;
             });

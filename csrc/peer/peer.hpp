@@ -86,6 +86,11 @@ class Peer {
     {
         return client_->egress_all();
     }
+    std::map<uint64_t, uint64_t> ingress_bytes() const
+    {
+        return server_ ? server_->ingress_all()
+                       : std::map<uint64_t, uint64_t>{};
+    }
 
   private:
     void update_to(const PeerList &workers, const PeerList &runners);

@@ -480,6 +480,16 @@ PYBIND11_MODULE(_core, m)
         return out;
     });
     m.def("trace_enabled", [] { return Tracer::get().enabled; });
+    m.def("ingress_bytes", [] {
+        std::map<std::string, uint64_t> out;
+        for (auto &kv : peer().ingress_bytes()) {
+            PeerID p;
+            p.ipv4 = (uint32_t)(kv.first >> 16);
+            p.port = (uint16_t)(kv.first & 0xffff);
+            out[p.str()] = kv.second;
+        }
+        return out;
+    });
     m.def("egress_bytes", [] {
         auto eg = peer().egress_bytes();
         py::dict d;

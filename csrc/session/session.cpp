@@ -4,6 +4,8 @@
 #include <cstring>
 #include <thread>
 
+#include "../core/workers.hpp"
+
 namespace kf {
 
 namespace {
@@ -76,6 +78,7 @@ void Session::run_graphs(const Workspace &w, const GraphPair &g,
         if (buf.size() != bytes)
             throw std::runtime_error("reduce size mismatch on " + rname);
         reduce_inplace(acc, buf.data(), w.count, w.dt, w.op);
+        collective_.recycle(std::move(buf));
     }
     for (int nx : rg.nexts[rank_]) {
         client_.send(peer(nx), ConnType::Collective, rname, 0, acc, bytes);
@@ -145,13 +148,20 @@ void Session::run_strategies(const Workspace &w,
     }
 
     const size_t per = (w.count + nchunks - 1) / nchunks;
-    std::vector<std::thread> threads;
     std::vector<std::string> errors(nchunks);
-    threads.reserve(nchunks);
+    size_t launched = 0;
     for (size_t c = 0; c < nchunks; ++c) {
         const size_t begin = c * per;
+        if (std::min(per, w.count - begin) == 0 || begin >= w.count) break;
+        ++launched;
+    }
+    Latch latch((int)launched);
+    auto &pool = CachedThreadPool::inst();
+    // chunks run on the cached worker pool (no thread spawn per chunk);
+    // chunk 0 runs inline on the caller thread
+    for (size_t c = 1; c < launched; ++c) {
+        const size_t begin = c * per;
         const size_t cnt = std::min(per, w.count - begin);
-        if (cnt == 0 || begin >= w.count) break;
         Workspace cw = w;
         cw.send = w.send ? (const uint8_t *)w.send + begin * elem : nullptr;
         cw.recv = (uint8_t *)w.recv + begin * elem;
@@ -160,16 +170,29 @@ void Session::run_strategies(const Workspace &w,
             (fnv1a(w.name) + c) % strategies.size();  // spread chunks
         std::string sfx = sfx_base + "." + std::to_string(c) + "@" +
                           std::to_string(sidx);
-        threads.emplace_back([this, cw, &strategies, sidx, sfx, &errors,
-                              c] {
+        pool.submit([this, cw, &strategies, sidx, sfx, &errors, c,
+                     &latch] {
             try {
                 run_graphs(cw, strategies[sidx], sfx);
             } catch (const std::exception &e) {
                 errors[c] = e.what();
             }
+            latch.done();
         });
     }
-    for (auto &t : threads) t.join();
+    if (launched > 0) {
+        Workspace cw = w;
+        cw.count = std::min(per, w.count);
+        const size_t sidx = fnv1a(w.name) % strategies.size();
+        try {
+            run_graphs(cw, strategies[sidx],
+                       sfx_base + ".0@" + std::to_string(sidx));
+        } catch (const std::exception &e) {
+            errors[0] = e.what();
+        }
+        latch.done();
+    }
+    latch.wait();
     for (auto &e : errors) {
         if (!e.empty()) throw std::runtime_error("chunk failed: " + e);
     }
@@ -220,13 +243,19 @@ void Session::all_gather(const Workspace &w)
     const int n = peers_.size();
     uint8_t *out = (uint8_t *)w.recv;
     std::memcpy(out + (size_t)rank_ * bytes, w.send, bytes);
-    // full-mesh exchange (reference allgather.go:17-45)
-    std::vector<std::thread> threads;
+    // full-mesh exchange (reference allgather.go:17-45); sends fan out on
+    // the cached worker pool
+    Latch latch(n - 1 > 0 ? n - 1 : 0);
+    auto &pool = CachedThreadPool::inst();
     for (int r = 0; r < n; ++r) {
         if (r == rank_) continue;
-        threads.emplace_back([this, r, name, bytes, &w] {
-            client_.send(peer(r), ConnType::Collective, name, 0, w.send,
-                         bytes);
+        pool.submit([this, r, name, bytes, &w, &latch] {
+            try {
+                client_.send(peer(r), ConnType::Collective, name, 0,
+                             w.send, bytes);
+            } catch (...) {
+            }
+            latch.done();
         });
     }
     for (int r = 0; r < n; ++r) {
@@ -234,7 +263,7 @@ void Session::all_gather(const Workspace &w)
         collective_.recv_into(peer(r), name, out + (size_t)r * bytes,
                               bytes);
     }
-    for (auto &t : threads) t.join();
+    latch.wait();
 }
 
 void Session::gather(const Workspace &w)
