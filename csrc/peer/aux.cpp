@@ -14,6 +14,7 @@
 
 #include <arpa/inet.h>
 #include <netinet/in.h>
+#include <cctype>
 #include <sched.h>
 #include <sys/socket.h>
 #include <unistd.h>
@@ -22,23 +23,103 @@
 #include <cstdio>
 #include <cstring>
 #include <sstream>
+#include <string>
 #include <thread>
+#include <vector>
 
 namespace kf {
+
+namespace {
+
+// Parse a kernel cpulist string like "0-31,64-95" into cpu indices.
+std::vector<int> parse_cpulist(const std::string &s)
+{
+    std::vector<int> cpus;
+    size_t i = 0;
+    while (i < s.size()) {
+        size_t j = s.find(',', i);
+        std::string part = s.substr(i, j == std::string::npos ? j : j - i);
+        size_t dash = part.find('-');
+        try {
+            if (dash == std::string::npos) {
+                if (!part.empty() && isdigit((unsigned char)part[0]))
+                    cpus.push_back(std::stoi(part));
+            } else {
+                int a = std::stoi(part.substr(0, dash));
+                int b = std::stoi(part.substr(dash + 1));
+                for (int c = a; c <= b; ++c) cpus.push_back(c);
+            }
+        } catch (...) {
+        }
+        if (j == std::string::npos) break;
+        i = j + 1;
+    }
+    return cpus;
+}
+
+// NUMA-domain CPU lists from sysfs, in node order (the reference uses
+// hwloc PU ordering, srcs/cpp/src/numa/affinity.cpp:26-63; sysfs gives
+// the same physical grouping without the library).
+std::vector<std::vector<int>> numa_nodes()
+{
+    std::vector<std::vector<int>> nodes;
+    for (int n = 0; n < 64; ++n) {
+        std::string path = "/sys/devices/system/node/node" +
+                           std::to_string(n) + "/cpulist";
+        FILE *f = std::fopen(path.c_str(), "r");
+        if (!f) break;
+        char buf[4096];
+        std::string s;
+        if (std::fgets(buf, sizeof(buf), f)) s = buf;
+        std::fclose(f);
+        auto cpus = parse_cpulist(s);
+        if (!cpus.empty()) nodes.push_back(std::move(cpus));
+    }
+    return nodes;
+}
+
+}  // namespace
 
 int bind_cpu_affinity(int local_rank, int local_size)
 {
     const long ncpu = sysconf(_SC_NPROCESSORS_ONLN);
     if (ncpu <= 0 || local_size <= 0) return -1;
-    const long per = ncpu / local_size > 0 ? ncpu / local_size : 1;
-    const long lo = (local_rank % local_size) * per;
     cpu_set_t set;
     CPU_ZERO(&set);
-    for (long c = lo; c < lo + per && c < ncpu; ++c) {
-        CPU_SET((int)c, &set);
+    // NUMA-aware placement: ranks are spread across NUMA domains and each
+    // rank's CPU slice stays INSIDE one domain (an even global partition
+    // interleaves wrongly across SMT/NUMA boundaries on a real MI355X
+    // host). Falls back to the even partition when sysfs has no topology.
+    const auto nodes = numa_nodes();
+    int bound = 0;
+    if (nodes.size() > 1) {
+        const int nn = (int)nodes.size();
+        const int node = local_rank % nn;
+        // ranks sharing a node split its cpulist evenly
+        const int ranks_here =
+            local_size / nn + (local_rank % nn < local_size % nn ? 1 : 0);
+        const int sub = local_rank / nn;  // index among ranks on this node
+        const auto &cpus = nodes[node];
+        const int per =
+            ranks_here > 0 ? (int)cpus.size() / ranks_here : (int)cpus.size();
+        if (per > 0) {
+            for (int i = sub * per;
+                 i < (sub + 1) * per && i < (int)cpus.size(); ++i) {
+                CPU_SET(cpus[i], &set);
+                ++bound;
+            }
+        }
+    }
+    if (bound == 0) {
+        const long per = ncpu / local_size > 0 ? ncpu / local_size : 1;
+        const long lo = (local_rank % local_size) * per;
+        for (long c = lo; c < lo + per && c < ncpu; ++c) {
+            CPU_SET((int)c, &set);
+            ++bound;
+        }
     }
     if (sched_setaffinity(0, sizeof(set), &set) != 0) return -1;
-    return (int)per;
+    return bound;
 }
 
 // ---------- StallDetector ----------

@@ -847,3 +847,56 @@ def rccl_gpu_pair_body(rank, np):
     rccl.finalize()
     kf.finalize()
     return "ok" if ok else "bad-sum"
+
+
+def ingress_bytes_body(rank, np):
+    import numpy as np_
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+
+    kf.init(with_torch=False)
+    x = np_.ones(200_000, dtype=np_.float32)
+    _core.all_reduce(x.ctypes.data, x.ctypes.data, x.size, 10, 0, "ing")
+    kf.barrier()
+    total_in = sum(_core.ingress_bytes().values())
+    kf.finalize()
+    return total_in > 0
+
+
+def rccl_cpu_elastic_reinit_body(rank, np):
+    """Elastic lifecycle through the native RCCL layer API on CPU: after a
+    resize, reinit() rebuilds every scope's rank/size from the NEW cluster
+    through the same control-plane capsule (reference ResetNcclHelper,
+    ops/gpu/scheduler.cpp:43-72) without any process-group teardown."""
+    import kungfu_amd as kf
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    rccl.init_cpu()
+    assert rccl.scope_size(rccl.GLOBAL) == np
+    # shrink to np-1: rank np-1 detaches
+    if rank == 0:
+        kf.propose_new_size(np - 1)
+    kf.barrier()
+    changed, detached = kf.resize()
+    if detached:
+        rccl.finalize()
+        kf.finalize()
+        return "detached"
+    assert changed
+    rccl.reinit()
+    assert rccl.scope_size(rccl.GLOBAL) == np - 1
+    assert rccl.scope_rank(rccl.GLOBAL) == rank
+    assert rccl.scope_size(rccl.LOCAL) == np - 1
+    # dispatcher still functional after reinit
+    from kungfu_amd import _rccl
+
+    got = []
+    rccl.scheduler_reset(["x", "y"])
+    _rccl.start_task(0, "y", lambda: got.append("y"))
+    _rccl.start_task(0, "x", lambda: got.append("x"))
+    _rccl.drain(0)
+    assert got == ["x", "y"]
+    rccl.finalize()
+    kf.finalize()
+    return "ok"

@@ -60,3 +60,26 @@ def test_kungfu_run_propagates_failure(port_block):
         sys.executable, "-c", "import sys; sys.exit(3)",
     ])
     assert r.returncode == 3
+
+
+def test_rrun_dry_run():
+    """kungfu-rrun builds one ssh command per WORKER with the full env
+    protocol (reference cmd/kungfu-rrun/rrun.go: workers launched
+    directly, no per-host launcher)."""
+    import subprocess
+    import sys
+
+    out = subprocess.run(
+        [sys.executable, "-m", "kungfu_amd.launcher.rrun", "-np", "4",
+         "-H", "10.0.0.1:2,10.0.0.2:2", "--dry-run", "--",
+         "python3", "train.py"],
+        capture_output=True, text=True, cwd=ROOT)
+    assert out.returncode == 0, out.stderr
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("[")]
+    assert len(lines) == 4
+    assert "KUNGFU_SELF_SPEC=10.0.0.1:30100" in lines[0]
+    assert "ssh" in lines[0] and "train.py" in lines[0]
+    # round-robin-by-slot: ranks alternate hosts
+    assert "10.0.0.2:30100" in lines[1]
+    assert "KUNGFU_INIT_PEERS" in lines[0]
+    assert "CUDA_VISIBLE_DEVICES=1" in lines[2]

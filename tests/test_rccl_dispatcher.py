@@ -44,3 +44,40 @@ def test_rccl_order_agreement(port_block):
 
 def test_rccl_local_scope_agreement(port_block):
     _run_local(port_block)
+
+
+@retry_flaky
+def _run_elastic(port_block):
+    from tests.mp_helpers import rccl_cpu_elastic_reinit_body
+
+    # needs a config server for resize: run under the launcher
+    import os
+    import re
+    import subprocess
+    import sys
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = ("import sys; sys.path.insert(0, %r); "
+              "from tests.mp_helpers import rccl_cpu_elastic_reinit_body; "
+              "import kungfu_amd as kf; kf.init(with_torch=False); "
+              "r = rccl_cpu_elastic_reinit_body(kf.rank(), kf.size()); "
+              "print('RESULT', r, flush=True)" % ROOT)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    from tests.mp_helpers import run_launcher_graceful
+
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "3", "-w",
+         "-port", str(port_block), "-port-range", str(port_block + 2),
+         "-builtin-config-port", str(port_block + 1),
+         sys.executable, "-c", script], ROOT, env, 120)
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
+    assert rc == 0, plain + err
+    oks = [ln for ln in plain.splitlines() if "RESULT ok" in ln]
+    # the detached worker may be killed by the watch runner before its
+    # print lands; survivors are what matters
+    assert len(oks) == 2, plain
+
+
+def test_rccl_elastic_reinit_cpu(port_block):
+    _run_elastic(port_block)

@@ -138,3 +138,12 @@ def test_cli_tools_exist():
     for t in ("kungfu-run", "kungfu-config-server", "kungfu-distribute"):
         p = os.path.join(ROOT, "tools", t)
         assert os.path.exists(p) and os.access(p, os.X_OK)
+
+
+def test_ingress_accounting(port_block):
+    """Both traffic directions are metered (reference monitor/counters.go;
+    round-1 gap: egress only)."""
+    from mp_helpers import ingress_bytes_body, spawn_cluster
+
+    res = spawn_cluster(ingress_bytes_body, 2, port_block)
+    assert res == [True, True]
