@@ -40,6 +40,7 @@ void Session::set_tree(const std::vector<int> &parent)
     if ((int)parent.size() != peers_.size())
         throw std::runtime_error("set_tree: bad forest size");
     global_ = {gen_from_forest(parent)};
+    reduce_only_.clear();
     std::lock_guard<std::mutex> lk(stats_mu_);
     stats_.assign(1, {});
 }
@@ -48,6 +49,7 @@ void Session::set_strategy(Strategy s)
 {
     strategy_kind_ = s;
     global_ = gen_strategies(peers_, s);
+    reduce_only_.clear();
     std::lock_guard<std::mutex> lk(stats_mu_);
     stats_.assign(global_.size(), {});
 }
@@ -214,14 +216,20 @@ void Session::all_reduce(const Workspace &w)
 
 void Session::reduce(const Workspace &w)
 {
-    // reduce-graph-only run: we reuse run_graphs but skip bcast by using a
-    // pair whose bcast graph has no edges except at the root.
-    const auto &g = global_[0];
-    Workspace ww = w;
-    GraphPair p;
-    p.reduce = g.reduce;
-    p.bcast = Graph(p.reduce.n);  // empty bcast: result stays at root
-    run_graphs(ww, p, "@rd");
+    // reduce-only strategy list (reduce graphs + empty bcast), built once
+    // from ALL global strategies so chunked reduces rotate across
+    // topologies exactly like all-reduce (round-1 gap: always used
+    // global_[0] and ran unchunked)
+    if (reduce_only_.size() != global_.size()) {
+        reduce_only_.clear();
+        for (const auto &g : global_) {
+            GraphPair p;
+            p.reduce = g.reduce;
+            p.bcast = Graph(p.reduce.n);  // result stays at root
+            reduce_only_.push_back(std::move(p));
+        }
+    }
+    run_strategies(w, reduce_only_, false);
 }
 
 void Session::broadcast(const Workspace &w, int root)
@@ -230,7 +238,16 @@ void Session::broadcast(const Workspace &w, int root)
     if (root == 0) {
         run_bcast_graph(w, global_[0].bcast, sfx);
     } else {
-        auto pair = gen_star(peers_.size(), root);
+        // cache per-root star graphs (round-1 gap: regenerated per call)
+        std::unique_lock<std::mutex> lk(stats_mu_);
+        auto it = root_bcast_.find(root);
+        if (it == root_bcast_.end()) {
+            it = root_bcast_
+                     .emplace(root, gen_star(peers_.size(), root))
+                     .first;
+        }
+        GraphPair pair = it->second;
+        lk.unlock();
         run_bcast_graph(w, pair.bcast, sfx);
     }
 }

@@ -14,6 +14,7 @@
 #pragma once
 
 #include <functional>
+#include <map>
 #include <string>
 #include <vector>
 
@@ -93,6 +94,8 @@ class Session {
     CollectiveEndpoint &collective_;
     Strategy strategy_kind_;
     std::vector<GraphPair> global_, local_, cross_;
+    std::vector<GraphPair> reduce_only_;      // built lazily from global_
+    std::map<int, GraphPair> root_bcast_;     // cached non-zero-root stars
     mutable std::mutex stats_mu_;
     std::vector<StrategyStat> stats_;
     double best_throughput_ = 0;
