@@ -1,11 +1,49 @@
 // Elementwise reduction for the CPU collective engine.
-// Reference parity: srcs/go/kungfu/base/op.cpp (std_transform_2) + f16.c.
-// Plain templated loops; the compiler auto-vectorizes the f32/i32 cases.
+// Reference parity: srcs/go/kungfu/base/op.cpp (std_transform_2) + f16.c
+// (AVX _mm256_cvtph_ps f16 sum). Plain templated loops (auto-vectorized
+// f32/i32) plus an F16C+AVX vector path for f16 sums when the host CPU
+// supports it.
 #include "common.hpp"
+
+#if defined(__x86_64__)
+#include <immintrin.h>
+#endif
 
 namespace kf {
 
 namespace {
+
+#if defined(__x86_64__)
+// f16 sum via F16C 8-lane convert-add-convert (reference f16.c:17-24),
+// gated on runtime CPU support so the binary stays portable.
+bool cpu_has_f16c()
+{
+    static const bool ok = __builtin_cpu_supports("f16c") &&
+                           __builtin_cpu_supports("avx");
+    return ok;
+}
+
+__attribute__((target("avx,f16c"))) void f16_sum_avx(uint16_t *acc,
+                                                     const uint16_t *in,
+                                                     size_t n)
+{
+    size_t i = 0;
+    for (; i + 8 <= n; i += 8) {
+        __m256 a = _mm256_cvtph_ps(
+            _mm_loadu_si128((const __m128i *)(acc + i)));
+        __m256 b = _mm256_cvtph_ps(
+            _mm_loadu_si128((const __m128i *)(in + i)));
+        _mm_storeu_si128(
+            (__m128i *)(acc + i),
+            _mm256_cvtps_ph(_mm256_add_ps(a, b),
+                            _MM_FROUND_TO_NEAREST_INT));
+    }
+    for (; i < n; ++i) {
+        acc[i] = float_to_half(half_to_float(acc[i]) +
+                               half_to_float(in[i]));
+    }
+}
+#endif
 
 template <typename T, typename F>
 void loop(T *acc, const T *in, size_t n, F f)
@@ -79,6 +117,12 @@ void reduce_inplace(void *acc, const void *in, size_t count, DType dt,
     case DType::U64:
         return dispatch_op((uint64_t *)acc, (const uint64_t *)in, count, op);
     case DType::F16:
+#if defined(__x86_64__)
+        if (op == ReduceOp::SUM && cpu_has_f16c()) {
+            return f16_sum_avx((uint16_t *)acc, (const uint16_t *)in,
+                               count);
+        }
+#endif
         return dispatch_f16ish<half_to_float, float_to_half>(
             (uint16_t *)acc, (const uint16_t *)in, count, op);
     case DType::BF16:
