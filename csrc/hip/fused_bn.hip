@@ -77,7 +77,21 @@ inline int bn_oct()
     return oct >= 2 ? 2 : 1;
 }
 
+// non-temporal loads for the single-use activation streams (KF_BN_NT=1)
+inline bool bn_nt()
+{
+    static const bool nt = env_int("KF_BN_NT", 0) != 0;
+    return nt;
+}
+
 typedef __attribute__((ext_vector_type(8))) unsigned short ushort8;
+
+template <bool NT>
+__device__ inline ushort8 load8(const unsigned short *p)
+{
+    if (NT) return __builtin_nontemporal_load((const ushort8 *)p);
+    return *(const ushort8 *)p;
+}
 
 __device__ inline float b2f(unsigned short u)
 {
@@ -103,7 +117,7 @@ __device__ inline unsigned short f2b(float f)
 // ---- pass 1: per-channel sum / sumsq ----
 // x: [M][C] bf16; out sums: f32[NSHADOW][2*C] (pre-zeroed)
 // OCT octets (8 channels / 16 B each) per lane; ILP rows in flight.
-template <int ILP, int OCT>
+template <int ILP, int OCT, bool NT>
 __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
                                 long long M, int C,
                                 float *__restrict__ sums)
@@ -128,7 +142,7 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
             for (int j = 0; j < ILP; ++j) {
 #pragma unroll
                 for (int o = 0; o < OCT; ++o)
-                    v[j][o] = *(const ushort8 *)(
+                    v[j][o] = load8<NT>(
                         x + (r + j * row_step) * C + g * 8 * OCT + o * 8);
             }
 #pragma unroll
@@ -147,8 +161,8 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
         for (; r < M; r += row_step) {
 #pragma unroll
             for (int o = 0; o < OCT; ++o) {
-                const ushort8 v = *(const ushort8 *)(x + r * C +
-                                                     g * 8 * OCT + o * 8);
+                const ushort8 v = load8<NT>(x + r * C + g * 8 * OCT +
+                                            o * 8);
 #pragma unroll
                 for (int k = 0; k < 8; ++k) {
                     const float f = b2f(v[k]);
@@ -173,7 +187,7 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
 // ---- finalize: mean/var -> folded scale/shift + running stats ----
 // sums: NSHADOW x {sum,sumsq}; outputs a = w*rstd, b = bias - mean*a;
 // saves mean/rstd for backward; updates running stats (momentum).
-__global__ void bn_finalize_kernel(const float *__restrict__ sums,
+__global__ void bn_finalize_kernel(float *__restrict__ sums,
                                    const float *__restrict__ weight,
                                    const float *__restrict__ bias,
                                    float *__restrict__ running_mean,
@@ -190,6 +204,9 @@ __global__ void bn_finalize_kernel(const float *__restrict__ sums,
     for (int k = 0; k < NSHADOW; ++k) {
         s0 += sums[(size_t)k * 2 * C + c];
         s1 += sums[(size_t)k * 2 * C + C + c];
+        // re-zero for the next step: kills the separate fill launch
+        sums[(size_t)k * 2 * C + c] = 0.f;
+        sums[(size_t)k * 2 * C + C + c] = 0.f;
     }
     const float mean = s0 / (float)M;
     const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
@@ -295,7 +312,7 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
 // dy_m = dy * relu_mask; xhat from save_mean/save_rstd.
 // out: f32[NSHADOW][2*C] {sum_dy, sum_dyxhat} (pre-zeroed; folded copies
 // ARE db and dw).
-template <int ILP, int OCT, bool MASKED>
+template <int ILP, int OCT, bool MASKED, bool NT>
 __global__ void bn_bwd_reduce_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
@@ -333,8 +350,8 @@ __global__ void bn_bwd_reduce_kernel(
                     (r + j * row_step) * C + (long long)g * 8 * OCT;
 #pragma unroll
                 for (int o = 0; o < OCT; ++o) {
-                    dv[j][o] = *(const ushort8 *)(dy + base + o * 8);
-                    xv[j][o] = *(const ushort8 *)(x + base + o * 8);
+                    dv[j][o] = load8<NT>(dy + base + o * 8);
+                    xv[j][o] = load8<NT>(x + base + o * 8);
                 }
                 if (MASKED) {
                     const unsigned char *mrow =
@@ -367,8 +384,8 @@ __global__ void bn_bwd_reduce_kernel(
             const long long base = r * C + (long long)g * 8 * OCT;
 #pragma unroll
             for (int o = 0; o < OCT; ++o) {
-                const ushort8 dv = *(const ushort8 *)(dy + base + o * 8);
-                const ushort8 xv = *(const ushort8 *)(x + base + o * 8);
+                const ushort8 dv = load8<NT>(dy + base + o * 8);
+                const ushort8 xv = load8<NT>(x + base + o * 8);
                 const unsigned char mb =
                     MASKED ? mask[r * (gpr * OCT) + g * OCT + o]
                            : (unsigned char)0xff;
@@ -396,14 +413,19 @@ __global__ void bn_bwd_reduce_kernel(
     }
 }
 
-// Fold the NSHADOW accumulator copies into copy 0.
-__global__ void bn_fold_kernel(float *__restrict__ sums, int C)
+// Fold the NSHADOW accumulator copies into dbdw[2*C] and re-zero the
+// shadows (no separate fill launch next step).
+__global__ void bn_fold_kernel(float *__restrict__ sums, int C,
+                               float *__restrict__ dbdw)
 {
     const int i = blockIdx.x * blockDim.x + threadIdx.x;
     if (i >= 2 * C) return;
-    float acc = sums[i];
-    for (int k = 1; k < NSHADOW; ++k) acc += sums[(size_t)k * 2 * C + i];
-    sums[i] = acc;
+    float acc = 0.f;
+    for (int k = 0; k < NSHADOW; ++k) {
+        acc += sums[(size_t)k * 2 * C + i];
+        sums[(size_t)k * 2 * C + i] = 0.f;
+    }
+    dbdw[i] = acc;
 }
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
@@ -511,28 +533,29 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const size_t lds = 2 * C * sizeof(float);
     const auto s = (hipStream_t)stream;
-#define CASE(I, O)                                                          \
-    hipLaunchKernelGGL((bn_stats_kernel<I, O>), grid, block, lds, s,        \
+#define CASE(I, O, NT)                                                      \
+    hipLaunchKernelGGL((bn_stats_kernel<I, O, NT>), grid, block, lds, s,    \
                        (const unsigned short *)x, M, C, (float *)sums)
+    const bool nt = bn_nt();
     if (bn_ilp_stats() == 4) {
-        if (oct == 2) CASE(4, 2);
-        else CASE(4, 1);
+        if (oct == 2) { if (nt) CASE(4, 2, true); else CASE(4, 2, false); }
+        else { if (nt) CASE(4, 1, true); else CASE(4, 1, false); }
     } else {
-        if (oct == 2) CASE(2, 2);
-        else CASE(2, 1);
+        if (oct == 2) { if (nt) CASE(2, 2, true); else CASE(2, 2, false); }
+        else { if (nt) CASE(2, 1, true); else CASE(2, 1, false); }
     }
 #undef CASE
     return hipGetLastError();
 }
 
-hipError_t kf_bn_finalize(const void *sums, const void *weight,
+hipError_t kf_bn_finalize(void *sums, const void *weight,
                           const void *bias, void *running_mean,
                           void *running_var, void *save_mean,
                           void *save_rstd, void *a, void *b, long long M,
                           int C, float eps, float momentum, void *stream)
 {
     hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
-                       0, (hipStream_t)stream, (const float *)sums,
+                       0, (hipStream_t)stream, (float *)sums,
                        (const float *)weight, (const float *)bias,
                        (float *)running_mean, (float *)running_var,
                        (float *)save_mean, (float *)save_rstd, (float *)a,
@@ -571,34 +594,37 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                             void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
-    const int oct = (C % 16 == 0) ? bn_oct() : 1;
+    const int oct = 1;  // OCT=2 measured 2x slower on every shape
     const int gpr = C / (8 * oct);
     const int rows_per_blk = BLOCK / gpr;
     const long long blocks = bn_reduce_blocks(M, C, rows_per_blk);
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
     const size_t lds = 2 * C * sizeof(float);
-#define CASE(I, O, MK)                                                      \
-    hipLaunchKernelGGL((bn_bwd_reduce_kernel<I, O, MK>), grid, block, lds, \
-                       s, (const unsigned short *)dy,                       \
+#define CASE(I, O, MK, NT)                                                  \
+    hipLaunchKernelGGL((bn_bwd_reduce_kernel<I, O, MK, NT>), grid, block,   \
+                       lds, s, (const unsigned short *)dy,                  \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)mean,    \
                        (const float *)rstd, M, C, (float *)sums)
+    const bool nt = bn_nt();
+    // OCT=2 measured 2x slower; only instantiate OCT=1 with the NT axis
     if (bn_ilp_bwd() == 4) {
-        if (oct == 2) { if (mask) CASE(4, 2, true); else CASE(4, 2, false); }
-        else { if (mask) CASE(4, 1, true); else CASE(4, 1, false); }
+        if (mask) { if (nt) CASE(4, 1, true, true); else CASE(4, 1, true, false); }
+        else { if (nt) CASE(4, 1, false, true); else CASE(4, 1, false, false); }
     } else {
-        if (oct == 2) { if (mask) CASE(2, 2, true); else CASE(2, 2, false); }
-        else { if (mask) CASE(2, 1, true); else CASE(2, 1, false); }
+        if (mask) { if (nt) CASE(2, 1, true, true); else CASE(2, 1, true, false); }
+        else { if (nt) CASE(2, 1, false, true); else CASE(2, 1, false, false); }
     }
 #undef CASE
     return hipGetLastError();
 }
 
-hipError_t kf_bn_fold(void *sums, int C, void *stream)
+hipError_t kf_bn_fold(void *sums, int C, void *dbdw, void *stream)
 {
     hipLaunchKernelGGL(bn_fold_kernel, dim3((2 * C + 255) / 256), dim3(256),
-                       0, (hipStream_t)stream, (float *)sums, C);
+                       0, (hipStream_t)stream, (float *)sums, C,
+                       (float *)dbdw);
     return hipGetLastError();
 }
 

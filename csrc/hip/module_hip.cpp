@@ -18,7 +18,7 @@ namespace py = pybind11;
 extern "C" {
 hipError_t kf_pack(const void *, int, void *, int, void *);
 hipError_t kf_bn_stats(const void *, long long, int, void *, void *);
-hipError_t kf_bn_finalize(const void *, const void *, const void *, void *,
+hipError_t kf_bn_finalize(void *, const void *, const void *, void *,
                           void *, void *, void *, void *, void *, long long,
                           int, float, float, void *);
 hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
@@ -26,7 +26,7 @@ hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
 hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, long long, int,
                             void *, void *);
-hipError_t kf_bn_fold(void *, int, void *);
+hipError_t kf_bn_fold(void *, int, void *, void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
                         const void *, long long, int, void *, void *,
@@ -250,7 +250,7 @@ PYBIND11_MODULE(_hip, m)
              uintptr_t b, long long M, int C, float eps, float momentum,
              uintptr_t stream) {
               check(kf_bn_finalize(
-                        (const void *)sums, (const void *)w,
+                        (void *)sums, (const void *)w,
                         (const void *)bias, (void *)rmean, (void *)rvar,
                         (void *)smean, (void *)srstd, (void *)a, (void *)b,
                         M, C, eps, momentum, (void *)stream),
@@ -277,9 +277,12 @@ PYBIND11_MODULE(_hip, m)
                                      (void *)sums, (void *)stream),
                     "kf_bn_bwd_reduce");
           });
-    m.def("bn_fold", [](uintptr_t sums, int C, uintptr_t stream) {
-        check(kf_bn_fold((void *)sums, C, (void *)stream), "kf_bn_fold");
-    });
+    m.def("bn_fold",
+          [](uintptr_t sums, int C, uintptr_t dbdw, uintptr_t stream) {
+              check(kf_bn_fold((void *)sums, C, (void *)dbdw,
+                               (void *)stream),
+                    "kf_bn_fold");
+          });
     m.def("bn_bwd_dx",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t a,
              uintptr_t mean, uintptr_t rstd, uintptr_t sums, long long M,

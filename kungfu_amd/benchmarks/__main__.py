@@ -31,7 +31,9 @@ def main():
     p.add_argument("--fuse", type=int, default=1)
     args = p.parse_args()
 
-    kf.init()
+    # CPU method must not bring up the GPU backend: on a box with fewer
+    # GPUs than ranks the RCCL bootstrap would fail on duplicate devices
+    kf.init(with_torch=args.method.startswith("RCCL"))
     sizes = model_sizes(args.model)
     dtype = torch.float32 if args.dtype == "f32" else torch.bfloat16
     use_gpu = args.method.startswith("RCCL")

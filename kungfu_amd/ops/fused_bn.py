@@ -39,8 +39,9 @@ class _FusedBNFunction(torch.autograd.Function):
         if training:
             # persistent per-module workspace: zero_() only, no per-call
             # allocations (was ~100 torch.zeros launches/step on ResNet-50)
+            # fwd_sums is zeroed by the PREVIOUS step's bn_finalize
+            # (zero-fused epilogue): no fill launch here
             sums = ws["fwd_sums"]
-            sums.zero_()
             _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)
             save_mean = ws["save_mean"]
             save_rstd = ws["save_rstd"]
@@ -84,21 +85,23 @@ class _FusedBNFunction(torch.autograd.Function):
         if dy.dtype != x.dtype:  # kernels require bf16 dy (matching x)
             dy = dy.to(x.dtype)
         dy = dy.contiguous(memory_format=torch.channels_last)
+        # bwd_sums was re-zeroed by the previous step's bn_fold
         sums = ctx.ws["bwd_sums"]
-        sums.zero_()
+        dbdw = ctx.ws["dbdw"]
         mask_ptr = mask.data_ptr() if ctx.has_mask else 0
         _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), mask_ptr,
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
                            C, sums.data_ptr(), s)
-        _hip.bn_fold(sums.data_ptr(), C, s)  # fold shadow accumulators
+        # fold shadows into dbdw and re-zero them (no fill next step)
+        _hip.bn_fold(sums.data_ptr(), C, dbdw.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), mask_ptr, a.data_ptr(),
                        save_mean.data_ptr(), save_rstd.data_ptr(),
-                       sums.data_ptr(), M, C, dx.data_ptr(),
+                       dbdw.data_ptr(), M, C, dx.data_ptr(),
                        dres.data_ptr() if dres is not None else 0, s)
-        db = sums[:C]               # db = sum(dy_m)
-        dw = sums[C:2 * C]          # dw = sum(dy_m * xhat)
+        db = dbdw[:C]               # db = sum(dy_m)
+        dw = dbdw[C:2 * C]          # dw = sum(dy_m * xhat)
         return (dx, dres, dw, db, None, None, None, None, None, None,
                 None)
 
@@ -130,6 +133,7 @@ class FusedBNReLU2d(torch.nn.Module):
             self._ws = {
                 "fwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
                 "bwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
+                "dbdw": torch.empty(2 * C, dtype=f32, device=dev),
                 "save_mean": torch.empty(C, dtype=f32, device=dev),
                 "save_rstd": torch.empty(C, dtype=f32, device=dev),
                 "a": torch.empty(C, dtype=f32, device=dev),
