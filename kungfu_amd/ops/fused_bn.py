@@ -85,15 +85,15 @@ class _FusedBNFunction(torch.autograd.Function):
         if dy.dtype != x.dtype:  # kernels require bf16 dy (matching x)
             dy = dy.to(x.dtype)
         dy = dy.contiguous(memory_format=torch.channels_last)
-        # bwd_sums was re-zeroed by the previous step's bn_fold
-        sums = ctx.ws["bwd_sums"]
+        # ONE launch: reduce + last-block fused fold into dbdw (shadows
+        # re-zeroed in-kernel for the next step)
         dbdw = ctx.ws["dbdw"]
         mask_ptr = mask.data_ptr() if ctx.has_mask else 0
         _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), mask_ptr,
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
-                           C, sums.data_ptr(), s)
-        # fold shadows into dbdw and re-zero them (no fill next step)
-        _hip.bn_fold(sums.data_ptr(), C, dbdw.data_ptr(), s)
+                           C, ctx.ws["bwd_sums"].data_ptr(),
+                           ctx.ws["ticket_bwd"].data_ptr(),
+                           dbdw.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), mask_ptr, a.data_ptr(),
@@ -134,6 +134,10 @@ class FusedBNReLU2d(torch.nn.Module):
                 "fwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
                 "bwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
                 "dbdw": torch.empty(2 * C, dtype=f32, device=dev),
+                "ticket_fwd": torch.zeros(1, dtype=torch.int32,
+                                          device=dev),
+                "ticket_bwd": torch.zeros(1, dtype=torch.int32,
+                                          device=dev),
                 "save_mean": torch.empty(C, dtype=f32, device=dev),
                 "save_rstd": torch.empty(C, dtype=f32, device=dev),
                 "a": torch.empty(C, dtype=f32, device=dev),
