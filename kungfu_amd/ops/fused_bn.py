@@ -37,22 +37,20 @@ class _FusedBNFunction(torch.autograd.Function):
         s = _stream()
         dev = x.device
         if training:
-            # persistent per-module workspace: zero_() only, no per-call
-            # allocations (was ~100 torch.zeros launches/step on ResNet-50)
-            # fwd_sums is zeroed by the PREVIOUS step's bn_finalize
-            # (zero-fused epilogue): no fill launch here
-            sums = ws["fwd_sums"]
-            _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)
+            # ONE launch: stats reduction + last-block fused finalize
+            # (shadow fold, a/b + saved stats, running-stat update,
+            # shadow re-zero). The persistent workspace means no per-call
+            # allocations and no fill launches at all.
             save_mean = ws["save_mean"]
             save_rstd = ws["save_rstd"]
             a = ws["a"]
             b = ws["b"]
-            _hip.bn_finalize(sums.data_ptr(), weight.data_ptr(),
-                             bias.data_ptr(), running_mean.data_ptr(),
-                             running_var.data_ptr(), save_mean.data_ptr(),
-                             save_rstd.data_ptr(), a.data_ptr(),
-                             b.data_ptr(), M, C, float(eps),
-                             float(momentum), s)
+            _hip.bn_stats(x.data_ptr(), M, C, ws["fwd_sums"].data_ptr(),
+                          ws["ticket_fwd"].data_ptr(), weight.data_ptr(),
+                          bias.data_ptr(), running_mean.data_ptr(),
+                          running_var.data_ptr(), save_mean.data_ptr(),
+                          save_rstd.data_ptr(), a.data_ptr(),
+                          b.data_ptr(), float(eps), float(momentum), s)
         else:
             rstd = torch.rsqrt(running_var + eps)
             a = (weight * rstd).float()
