@@ -50,29 +50,20 @@ def main():
     from kungfu_amd import _hip
 
     s = torch.cuda.current_stream().cuda_stream
-    f32, dev = torch.float32, "cuda"
-    sums = torch.zeros(16 * C, dtype=f32, device=dev)
-    tk = torch.zeros(1, dtype=torch.int32, device=dev)
-    ones = torch.ones(C, dtype=f32, device=dev)
-    zeros = torch.zeros(C, dtype=f32, device=dev)
-    scratch = [torch.empty(C, dtype=f32, device=dev) for _ in range(4)]
-    dbdw = torch.empty(2 * C, dtype=f32, device=dev)
+    sums = torch.zeros(16 * C, dtype=torch.float32, device="cuda")
+    mean = torch.zeros(C, dtype=torch.float32, device="cuda")
+    rstd = torch.ones(C, dtype=torch.float32, device="cuda")
     mask = torch.full((M * (C // 8),), 255, dtype=torch.uint8,
                       device="cuda")
     show("bn_stats (R)",
-         t(lambda: _hip.bn_stats(
-             x.data_ptr(), M, C, sums.data_ptr(), tk.data_ptr(),
-             ones.data_ptr(), zeros.data_ptr(), zeros.data_ptr(),
-             ones.data_ptr(), scratch[0].data_ptr(),
-             scratch[1].data_ptr(), scratch[2].data_ptr(),
-             scratch[3].data_ptr(), 1e-5, 0.0, s)), gb)
+         t(lambda: _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)),
+         gb)
     dy = y
     show("bn_bwd_reduce (2R)",
          t(lambda: _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(),
-                                      mask.data_ptr(), zeros.data_ptr(),
-                                      ones.data_ptr(), M, C,
-                                      sums.data_ptr(), tk.data_ptr(),
-                                      dbdw.data_ptr(), s)), 2 * gb)
+                                      mask.data_ptr(), mean.data_ptr(),
+                                      rstd.data_ptr(), M, C,
+                                      sums.data_ptr(), s)), 2 * gb)
 
 
 if __name__ == "__main__":

@@ -117,60 +117,10 @@ __device__ inline unsigned short f2b(float f)
 // ---- pass 1: per-channel sum / sumsq ----
 // x: [M][C] bf16; out sums: f32[NSHADOW][2*C] (pre-zeroed)
 // OCT octets (8 channels / 16 B each) per lane; ILP rows in flight.
-__device__ inline void bn_finalize_body(float *__restrict__ sums,
-                                        const float *__restrict__ weight,
-                                        const float *__restrict__ bias,
-                                        float *__restrict__ running_mean,
-                                        float *__restrict__ running_var,
-                                        float *__restrict__ save_mean,
-                                        float *__restrict__ save_rstd,
-                                        float *__restrict__ a,
-                                        float *__restrict__ b, long long M,
-                                        int C, float eps, float momentum)
-{
-    for (int c = threadIdx.x; c < C; c += BLOCK) {
-        float s0 = 0.f, s1 = 0.f;
-        for (int k = 0; k < NSHADOW; ++k) {
-            s0 += sums[(size_t)k * 2 * C + c];
-            s1 += sums[(size_t)k * 2 * C + C + c];
-            sums[(size_t)k * 2 * C + c] = 0.f;
-            sums[(size_t)k * 2 * C + C + c] = 0.f;
-        }
-        const float mean = s0 / (float)M;
-        const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
-        const float rstd = rsqrtf(var + eps);
-        save_mean[c] = mean;
-        save_rstd[c] = rstd;
-        const float ac = weight[c] * rstd;
-        a[c] = ac;
-        b[c] = bias[c] - mean * ac;
-        if (momentum > 0.f) {
-            running_mean[c] += momentum * (mean - running_mean[c]);
-            const float unbiased =
-                M > 1 ? var * (float)M / (float)(M - 1) : var;
-            running_var[c] += momentum * (unbiased - running_var[c]);
-        }
-    }
-}
-
-// stats + (last block) finalize in ONE launch: the block that finishes
-// last folds the shadow accumulators, computes a/b/save stats, updates
-// running stats, re-zeroes the shadows and resets the ticket — removing
-// one tiny kernel launch per BN call (53/step on ResNet-50).
 template <int ILP, int OCT, bool NT>
 __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
                                 long long M, int C,
-                                float *__restrict__ sums,
-                                unsigned int *__restrict__ ticket,
-                                const float *__restrict__ weight,
-                                const float *__restrict__ bias,
-                                float *__restrict__ running_mean,
-                                float *__restrict__ running_var,
-                                float *__restrict__ save_mean,
-                                float *__restrict__ save_rstd,
-                                float *__restrict__ a,
-                                float *__restrict__ b, float eps,
-                                float momentum)
+                                float *__restrict__ sums)
 {
     extern __shared__ float lds[];  // 2*C floats
     const int gpr = C / (8 * OCT);  // channel groups per row
@@ -232,17 +182,45 @@ __global__ void bn_stats_kernel(const unsigned short *__restrict__ x,
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
         atomicAdd(&shadow[i], lds[i]);
     }
-    // last-block fused finalize
-    __threadfence();
-    __shared__ unsigned int is_last;
-    __syncthreads();
-    if (threadIdx.x == 0)
-        is_last = (atomicAdd(ticket, 1u) == gridDim.x - 1) ? 1u : 0u;
-    __syncthreads();
-    if (is_last) {
-        bn_finalize_body(sums, weight, bias, running_mean, running_var,
-                         save_mean, save_rstd, a, b, M, C, eps, momentum);
-        if (threadIdx.x == 0) *ticket = 0;
+}
+
+// ---- finalize: mean/var -> folded scale/shift + running stats ----
+// sums: NSHADOW x {sum,sumsq}; outputs a = w*rstd, b = bias - mean*a;
+// saves mean/rstd for backward; updates running stats (momentum).
+__global__ void bn_finalize_kernel(float *__restrict__ sums,
+                                   const float *__restrict__ weight,
+                                   const float *__restrict__ bias,
+                                   float *__restrict__ running_mean,
+                                   float *__restrict__ running_var,
+                                   float *__restrict__ save_mean,
+                                   float *__restrict__ save_rstd,
+                                   float *__restrict__ a,
+                                   float *__restrict__ b, long long M,
+                                   int C, float eps, float momentum)
+{
+    const int c = blockIdx.x * blockDim.x + threadIdx.x;
+    if (c >= C) return;
+    float s0 = 0.f, s1 = 0.f;
+    for (int k = 0; k < NSHADOW; ++k) {
+        s0 += sums[(size_t)k * 2 * C + c];
+        s1 += sums[(size_t)k * 2 * C + C + c];
+        // re-zero for the next step: kills the separate fill launch
+        sums[(size_t)k * 2 * C + c] = 0.f;
+        sums[(size_t)k * 2 * C + C + c] = 0.f;
+    }
+    const float mean = s0 / (float)M;
+    const float var = fmaxf(s1 / (float)M - mean * mean, 0.f);
+    const float rstd = rsqrtf(var + eps);
+    save_mean[c] = mean;
+    save_rstd[c] = rstd;
+    const float ac = weight[c] * rstd;
+    a[c] = ac;
+    b[c] = bias[c] - mean * ac;
+    if (momentum > 0.f) {
+        running_mean[c] += momentum * (mean - running_mean[c]);
+        const float unbiased =
+            M > 1 ? var * (float)M / (float)(M - 1) : var;
+        running_var[c] += momentum * (unbiased - running_var[c]);
     }
 }
 
@@ -340,8 +318,7 @@ __global__ void bn_bwd_reduce_kernel(
     const unsigned short *__restrict__ x,
     const unsigned char *__restrict__ mask,
     const float *__restrict__ mean, const float *__restrict__ rstd,
-    long long M, int C, float *__restrict__ sums,
-    unsigned int *__restrict__ ticket, float *__restrict__ dbdw)
+    long long M, int C, float *__restrict__ sums)
 {
     extern __shared__ float lds[];  // 2*C floats
     const int gpr = C / (8 * OCT);
@@ -434,24 +411,21 @@ __global__ void bn_bwd_reduce_kernel(
     for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
         atomicAdd(&shadow[i], lds[i]);
     }
-    // last-block fused fold into dbdw (+ shadow re-zero)
-    __threadfence();
-    __shared__ unsigned int is_last;
-    __syncthreads();
-    if (threadIdx.x == 0)
-        is_last = (atomicAdd(ticket, 1u) == gridDim.x - 1) ? 1u : 0u;
-    __syncthreads();
-    if (is_last) {
-        for (int i = threadIdx.x; i < 2 * C; i += BLOCK) {
-            float acc = 0.f;
-            for (int k = 0; k < NSHADOW; ++k) {
-                acc += sums[(size_t)k * 2 * C + i];
-                sums[(size_t)k * 2 * C + i] = 0.f;
-            }
-            dbdw[i] = acc;
-        }
-        if (threadIdx.x == 0) *ticket = 0;
+}
+
+// Fold the NSHADOW accumulator copies into dbdw[2*C] and re-zero the
+// shadows (no separate fill launch next step).
+__global__ void bn_fold_kernel(float *__restrict__ sums, int C,
+                               float *__restrict__ dbdw)
+{
+    const int i = blockIdx.x * blockDim.x + threadIdx.x;
+    if (i >= 2 * C) return;
+    float acc = 0.f;
+    for (int k = 0; k < NSHADOW; ++k) {
+        acc += sums[(size_t)k * 2 * C + i];
+        sums[(size_t)k * 2 * C + i] = 0.f;
     }
+    dbdw[i] = acc;
 }
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
@@ -549,10 +523,7 @@ __global__ void bn_bwd_dx_kernel(
 extern "C" {
 
 hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
-                       void *ticket, const void *weight, const void *bias,
-                       void *running_mean, void *running_var,
-                       void *save_mean, void *save_rstd, void *a, void *b,
-                       float eps, float momentum, void *stream)
+                       void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int oct = (C % 16 == 0) ? bn_oct() : 1;
@@ -564,12 +535,7 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
     const auto s = (hipStream_t)stream;
 #define CASE(I, O, NT)                                                      \
     hipLaunchKernelGGL((bn_stats_kernel<I, O, NT>), grid, block, lds, s,    \
-                       (const unsigned short *)x, M, C, (float *)sums,      \
-                       (unsigned int *)ticket, (const float *)weight,       \
-                       (const float *)bias, (float *)running_mean,          \
-                       (float *)running_var, (float *)save_mean,            \
-                       (float *)save_rstd, (float *)a, (float *)b, eps,     \
-                       momentum)
+                       (const unsigned short *)x, M, C, (float *)sums)
     const bool nt = bn_nt();
     if (bn_ilp_stats() == 4) {
         if (oct == 2) { if (nt) CASE(4, 2, true); else CASE(4, 2, false); }
@@ -579,6 +545,21 @@ hipError_t kf_bn_stats(const void *x, long long M, int C, void *sums,
         else { if (nt) CASE(2, 1, true); else CASE(2, 1, false); }
     }
 #undef CASE
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_finalize(void *sums, const void *weight,
+                          const void *bias, void *running_mean,
+                          void *running_var, void *save_mean,
+                          void *save_rstd, void *a, void *b, long long M,
+                          int C, float eps, float momentum, void *stream)
+{
+    hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (float *)sums,
+                       (const float *)weight, (const float *)bias,
+                       (float *)running_mean, (float *)running_var,
+                       (float *)save_mean, (float *)save_rstd, (float *)a,
+                       (float *)b, M, C, eps, momentum);
     return hipGetLastError();
 }
 
@@ -610,8 +591,7 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
 hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                             const void *mask, const void *mean,
                             const void *rstd, long long M, int C,
-                            void *sums, void *ticket, void *dbdw,
-                            void *stream)
+                            void *sums, void *stream)
 {
     if (C % 8 != 0 || C / 8 > BLOCK) return hipErrorInvalidValue;
     const int oct = 1;  // OCT=2 measured 2x slower on every shape
@@ -626,8 +606,7 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
                        lds, s, (const unsigned short *)dy,                  \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)mean,    \
-                       (const float *)rstd, M, C, (float *)sums,            \
-                       (unsigned int *)ticket, (float *)dbdw)
+                       (const float *)rstd, M, C, (float *)sums)
     const bool nt = bn_nt();
     // OCT=2 measured 2x slower; only instantiate OCT=1 with the NT axis
     if (bn_ilp_bwd() == 4) {
@@ -638,6 +617,14 @@ hipError_t kf_bn_bwd_reduce(const void *dy, const void *x,
         else { if (nt) CASE(2, 1, false, true); else CASE(2, 1, false, false); }
     }
 #undef CASE
+    return hipGetLastError();
+}
+
+hipError_t kf_bn_fold(void *sums, int C, void *dbdw, void *stream)
+{
+    hipLaunchKernelGGL(bn_fold_kernel, dim3((2 * C + 255) / 256), dim3(256),
+                       0, (hipStream_t)stream, (float *)sums, C,
+                       (float *)dbdw);
     return hipGetLastError();
 }
 

@@ -17,14 +17,16 @@ namespace py = pybind11;
 
 extern "C" {
 hipError_t kf_pack(const void *, int, void *, int, void *);
-hipError_t kf_bn_stats(const void *, long long, int, void *, void *,
-                       const void *, const void *, void *, void *, void *,
-                       void *, void *, void *, float, float, void *);
+hipError_t kf_bn_stats(const void *, long long, int, void *, void *);
+hipError_t kf_bn_finalize(void *, const void *, const void *, void *,
+                          void *, void *, void *, void *, void *, long long,
+                          int, float, float, void *);
 hipError_t kf_bn_fwd(const void *, const void *, void *, const void *,
                      const void *, long long, int, int, void *, void *);
 hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, long long, int,
-                            void *, void *, void *, void *);
+                            void *, void *);
+hipError_t kf_bn_fold(void *, int, void *, void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
                         const void *, long long, int, void *, void *,
@@ -237,17 +239,22 @@ PYBIND11_MODULE(_hip, m)
     // ---- fused BatchNorm(+residual+ReLU), NHWC bf16 ----
     m.def("bn_stats",
           [](uintptr_t x, long long M, int C, uintptr_t sums,
-             uintptr_t ticket, uintptr_t w, uintptr_t bias,
-             uintptr_t rmean, uintptr_t rvar, uintptr_t smean,
-             uintptr_t srstd, uintptr_t a, uintptr_t b, float eps,
-             float momentum, uintptr_t stream) {
+             uintptr_t stream) {
               check(kf_bn_stats((const void *)x, M, C, (void *)sums,
-                                (void *)ticket, (const void *)w,
-                                (const void *)bias, (void *)rmean,
-                                (void *)rvar, (void *)smean,
-                                (void *)srstd, (void *)a, (void *)b, eps,
-                                momentum, (void *)stream),
+                                (void *)stream),
                     "kf_bn_stats");
+          });
+    m.def("bn_finalize",
+          [](uintptr_t sums, uintptr_t w, uintptr_t bias, uintptr_t rmean,
+             uintptr_t rvar, uintptr_t smean, uintptr_t srstd, uintptr_t a,
+             uintptr_t b, long long M, int C, float eps, float momentum,
+             uintptr_t stream) {
+              check(kf_bn_finalize(
+                        (void *)sums, (const void *)w,
+                        (const void *)bias, (void *)rmean, (void *)rvar,
+                        (void *)smean, (void *)srstd, (void *)a, (void *)b,
+                        M, C, eps, momentum, (void *)stream),
+                    "kf_bn_finalize");
           });
     m.def("bn_fwd",
           [](uintptr_t x, uintptr_t res, uintptr_t y, uintptr_t a,
@@ -262,14 +269,19 @@ PYBIND11_MODULE(_hip, m)
     m.def("bn_bwd_reduce",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t mean,
              uintptr_t rstd, long long M, int C, uintptr_t sums,
-             uintptr_t ticket, uintptr_t dbdw, uintptr_t stream) {
+             uintptr_t stream) {
               check(kf_bn_bwd_reduce((const void *)dy, (const void *)x,
                                      (const void *)mask,
                                      (const void *)mean,
                                      (const void *)rstd, M, C,
-                                     (void *)sums, (void *)ticket,
-                                     (void *)dbdw, (void *)stream),
+                                     (void *)sums, (void *)stream),
                     "kf_bn_bwd_reduce");
+          });
+    m.def("bn_fold",
+          [](uintptr_t sums, int C, uintptr_t dbdw, uintptr_t stream) {
+              check(kf_bn_fold((void *)sums, C, (void *)dbdw,
+                               (void *)stream),
+                    "kf_bn_fold");
           });
     m.def("bn_bwd_dx",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t a,

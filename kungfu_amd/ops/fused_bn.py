@@ -37,20 +37,22 @@ class _FusedBNFunction(torch.autograd.Function):
         s = _stream()
         dev = x.device
         if training:
-            # ONE launch: stats reduction + last-block fused finalize
-            # (shadow fold, a/b + saved stats, running-stat update,
-            # shadow re-zero). The persistent workspace means no per-call
-            # allocations and no fill launches at all.
+            # persistent per-module workspace: zero_() only, no per-call
+            # allocations (was ~100 torch.zeros launches/step on ResNet-50)
+            # fwd_sums is zeroed by the PREVIOUS step's bn_finalize
+            # (zero-fused epilogue): no fill launch here
+            sums = ws["fwd_sums"]
+            _hip.bn_stats(x.data_ptr(), M, C, sums.data_ptr(), s)
             save_mean = ws["save_mean"]
             save_rstd = ws["save_rstd"]
             a = ws["a"]
             b = ws["b"]
-            _hip.bn_stats(x.data_ptr(), M, C, ws["fwd_sums"].data_ptr(),
-                          ws["ticket_fwd"].data_ptr(), weight.data_ptr(),
-                          bias.data_ptr(), running_mean.data_ptr(),
-                          running_var.data_ptr(), save_mean.data_ptr(),
-                          save_rstd.data_ptr(), a.data_ptr(),
-                          b.data_ptr(), float(eps), float(momentum), s)
+            _hip.bn_finalize(sums.data_ptr(), weight.data_ptr(),
+                             bias.data_ptr(), running_mean.data_ptr(),
+                             running_var.data_ptr(), save_mean.data_ptr(),
+                             save_rstd.data_ptr(), a.data_ptr(),
+                             b.data_ptr(), M, C, float(eps),
+                             float(momentum), s)
         else:
             rstd = torch.rsqrt(running_var + eps)
             a = (weight * rstd).float()
@@ -83,15 +85,15 @@ class _FusedBNFunction(torch.autograd.Function):
         if dy.dtype != x.dtype:  # kernels require bf16 dy (matching x)
             dy = dy.to(x.dtype)
         dy = dy.contiguous(memory_format=torch.channels_last)
-        # ONE launch: reduce + last-block fused fold into dbdw (shadows
-        # re-zeroed in-kernel for the next step)
+        # bwd_sums was re-zeroed by the previous step's bn_fold
+        sums = ctx.ws["bwd_sums"]
         dbdw = ctx.ws["dbdw"]
         mask_ptr = mask.data_ptr() if ctx.has_mask else 0
         _hip.bn_bwd_reduce(dy.data_ptr(), x.data_ptr(), mask_ptr,
                            save_mean.data_ptr(), save_rstd.data_ptr(), M,
-                           C, ctx.ws["bwd_sums"].data_ptr(),
-                           ctx.ws["ticket_bwd"].data_ptr(),
-                           dbdw.data_ptr(), s)
+                           C, sums.data_ptr(), s)
+        # fold shadows into dbdw and re-zero them (no fill next step)
+        _hip.bn_fold(sums.data_ptr(), C, dbdw.data_ptr(), s)
         dx = torch.empty_like(x)
         dres = torch.empty_like(x) if ctx.has_res else None
         _hip.bn_bwd_dx(dy.data_ptr(), x.data_ptr(), mask_ptr, a.data_ptr(),
@@ -132,10 +134,6 @@ class FusedBNReLU2d(torch.nn.Module):
                 "fwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
                 "bwd_sums": torch.zeros(16 * C, dtype=f32, device=dev),
                 "dbdw": torch.empty(2 * C, dtype=f32, device=dev),
-                "ticket_fwd": torch.zeros(1, dtype=torch.int32,
-                                          device=dev),
-                "ticket_bwd": torch.zeros(1, dtype=torch.int32,
-                                          device=dev),
                 "save_mean": torch.empty(C, dtype=f32, device=dev),
                 "save_rstd": torch.empty(C, dtype=f32, device=dev),
                 "a": torch.empty(C, dtype=f32, device=dev),
