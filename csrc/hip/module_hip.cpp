@@ -76,33 +76,18 @@ class FusionPlan {
     FusionPlan(const std::vector<std::tuple<uintptr_t, uint64_t, uint64_t>>
                    &segments,
                int dtype, unsigned chunk_elems = 1u << 14)
-        : dtype_(dtype)
+        : dtype_(dtype), chunk_elems_(chunk_elems)
     {
-        std::vector<ChunkHost> chunks;
-        total_ = 0;
-        for (const auto &seg : segments) {
-            const uintptr_t ptr = std::get<0>(seg);
-            const uint64_t off = std::get<1>(seg);
-            uint64_t n = std::get<2>(seg);
-            total_ += n;
-            uint64_t done = 0;
-            const size_t esize = dtype == 0 ? 4 : 2;
-            while (done < n) {
-                const uint64_t take = std::min<uint64_t>(chunk_elems,
-                                                         n - done);
-                ChunkHost c;
-                c.src = (const void *)(ptr + done * esize);
-                c.dst = off + done;
-                c.n = (unsigned int)take;
-                chunks.push_back(c);
-                done += take;
-            }
-        }
-        nchunks_ = (int)chunks.size();
+        build(segments);
+        cap_ = nchunks_;
         if (nchunks_ > 0) {
             check(hipMalloc(&dev_, sizeof(ChunkHost) * nchunks_),
                   "hipMalloc(FusionPlan)");
-            check(hipMemcpy(dev_, chunks.data(),
+            check(hipHostMalloc(&host_, sizeof(ChunkHost) * nchunks_, 0),
+                  "hipHostMalloc(FusionPlan)");
+            std::memcpy(host_, chunks_.data(),
+                        sizeof(ChunkHost) * nchunks_);
+            check(hipMemcpy(dev_, chunks_.data(),
                             sizeof(ChunkHost) * nchunks_,
                             hipMemcpyHostToDevice),
                   "hipMemcpy(FusionPlan)");
@@ -111,8 +96,26 @@ class FusionPlan {
     ~FusionPlan()
     {
         if (dev_) (void)hipFree(dev_);
+        if (host_) (void)hipHostFree(host_);
     }
     FusionPlan(const FusionPlan &) = delete;
+
+    // Re-point the chunk table at new source buffers (same segment
+    // lengths/offsets layout): async H2D from pinned staging on `stream`.
+    // Used by the pack-mode gradient path, where autograd's grad tensors
+    // get fresh addresses every step.
+    void update(const std::vector<std::tuple<uintptr_t, uint64_t,
+                                             uint64_t>> &segments,
+                uintptr_t stream)
+    {
+        build(segments);
+        if (nchunks_ > cap_)
+            throw std::runtime_error("FusionPlan.update: table grew");
+        std::memcpy(host_, chunks_.data(), sizeof(ChunkHost) * nchunks_);
+        check(hipMemcpyAsync(dev_, host_, sizeof(ChunkHost) * nchunks_,
+                             hipMemcpyHostToDevice, (hipStream_t)stream),
+              "hipMemcpyAsync(FusionPlan.update)");
+    }
 
     void pack(uintptr_t fused, uintptr_t stream)
     {
@@ -130,6 +133,36 @@ class FusionPlan {
     int nchunks() const { return nchunks_; }
 
   private:
+    void build(const std::vector<std::tuple<uintptr_t, uint64_t,
+                                            uint64_t>> &segments)
+    {
+        chunks_.clear();
+        total_ = 0;
+        for (const auto &seg : segments) {
+            const uintptr_t ptr = std::get<0>(seg);
+            const uint64_t off = std::get<1>(seg);
+            uint64_t n = std::get<2>(seg);
+            total_ += n;
+            uint64_t done = 0;
+            const size_t esize = dtype_ == 0 ? 4 : 2;
+            while (done < n) {
+                const uint64_t take =
+                    std::min<uint64_t>(chunk_elems_, n - done);
+                ChunkHost c;
+                c.src = (const void *)(ptr + done * esize);
+                c.dst = off + done;
+                c.n = (unsigned int)take;
+                chunks_.push_back(c);
+                done += take;
+            }
+        }
+        nchunks_ = (int)chunks_.size();
+    }
+
+    std::vector<ChunkHost> chunks_;
+    unsigned chunk_elems_;
+    int cap_ = 0;
+    void *host_ = nullptr;
     int dtype_;
     int nchunks_ = 0;
     void *dev_ = nullptr;
@@ -150,6 +183,8 @@ PYBIND11_MODULE(_hip, m)
              py::arg("segments"), py::arg("dtype"),
              py::arg("chunk_elems") = 1u << 14)
         .def("pack", &FusionPlan::pack, py::arg("fused"), py::arg("stream"))
+        .def("update", &FusionPlan::update, py::arg("segments"),
+             py::arg("stream"))
         .def("unpack", &FusionPlan::unpack, py::arg("fused"),
              py::arg("scale"), py::arg("stream"))
         .def_property_readonly("total_elems", &FusionPlan::total_elems)

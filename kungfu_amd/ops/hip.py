@@ -78,7 +78,17 @@ class FusionPlan:
         ptrs = [t.data_ptr() for t in self._tensors]
         if ptrs != self._ptrs:
             self._ptrs = ptrs
-            self._plan = self._build()
+            segs = [(t.data_ptr(), off, t.numel())
+                    for t, off in zip(self._tensors, self._offsets)]
+            # async table re-point (pinned staging): safe per-step cost
+            self._plan.update(segs, _stream())
+
+    def set_sources(self, tensors):
+        """Re-point the plan at new source tensors with the same segment
+        layout (pack-mode gradients: autograd allocates fresh grad
+        tensors every step)."""
+        self._tensors = list(tensors)
+        self._revalidate()
 
     def pack(self, fused):
         _check_cuda(fused)

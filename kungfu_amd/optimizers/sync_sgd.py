@@ -17,10 +17,16 @@ class SynchronousSGDOptimizer(KungFuOptimizer):
     def __init__(self, optimizer, bucket_bytes=DEFAULT_BUCKET_BYTES,
                  overlap=True, average=True, name="sgd", fused_step=False):
         super().__init__(optimizer)
+        # fused step -> pack-mode gradients: autograd keeps its own grad
+        # tensors (first accumulation is a steal, no per-param zero/add
+        # kernels) and one HIP pack per bucket feeds the flat buffers the
+        # fused optimizer consumes
         self.reducer = GradBucketReducer(self._params(),
                                          bucket_bytes=bucket_bytes,
                                          average=average, overlap=overlap,
-                                         name=name)
+                                         name=name,
+                                         mode="pack" if fused_step
+                                         else "alias")
         self.fused_step = fused_step
         if fused_step:
             self._build_fused_step()

@@ -65,7 +65,8 @@ class _NativeBucketWork:
 
 class _Bucket:
     __slots__ = ("index", "params", "flat", "numel", "ready", "work",
-                 "launched", "param_flat", "momentum", "master")
+                 "launched", "param_flat", "momentum", "master",
+                 "offsets", "plan")
 
     def __init__(self, index):
         self.index = index
@@ -79,8 +80,18 @@ class _Bucket:
 
 
 class GradBucketReducer:
+    """mode="alias": param.grad is a view into the flat bucket (autograd
+    accumulates straight into the all-reduce buffer; costs a zero-fill +
+    read-modify-write add per param per step).
+    mode="pack": autograd keeps its own grad tensors (first accumulation
+    is a zero-cost steal) and a single HIP pack kernel per bucket gathers
+    them into the flat buffer — no per-param zero/add kernels at all
+    (~165 tiny launches/step on ResNet-50). Requires the fused optimizer
+    (param updates read the flat buffers, reduced values never go back
+    into p.grad)."""
+
     def __init__(self, params, bucket_bytes=DEFAULT_BUCKET_BYTES,
-                 average=True, overlap=True, name="grads"):
+                 average=True, overlap=True, name="grads", mode="alias"):
         _ensure_init()
         self.params = [p for p in params if p.requires_grad]
         if not self.params:
@@ -88,13 +99,20 @@ class GradBucketReducer:
         self.average = average
         self.overlap = overlap
         self.name = name
+        self.mode = mode
         self.world = _core.size()
         dev = self.params[0].device
         self.is_cuda = dev.type == "cuda"
+        if mode == "pack":
+            from kungfu_amd.ops import hip as hip_ops
+
+            if not (self.is_cuda and hip_ops.available()):
+                self.mode = mode = "alias"  # CPU / no-HIP fallback
         self._build_buckets(bucket_bytes)
         self._hooks = []
         self._next_launch = 0
-        if self.overlap and self.is_cuda and self.world > 1:
+        if self.is_cuda and (self.world > 1 or self.mode == "pack") \
+                and self.overlap:
             self._register_hooks()
 
     # -- construction ---------------------------------------------------
@@ -129,23 +147,27 @@ class GradBucketReducer:
         for b in self.buckets:
             if b.index not in self._launch_seq:
                 self._launch_seq.append(b.index)
-        # allocate flats and alias grads
+        # allocate flats; alias grads (alias mode) or build pack plans
         self.bucket_of = {}
         for b in self.buckets:
             dtype = b.params[0].dtype
             dev = b.params[0].device
             b.flat = torch.zeros(b.numel, dtype=dtype, device=dev)
             off = 0
+            b.offsets = []
             for p in b.params:
                 if p.dtype != dtype or p.device != dev:
                     raise ValueError(
                         "mixed dtype/device parameters in one reducer")
-                view = _alias_view(b.flat[off:off + p.numel()], p)
-                if p.grad is not None:
-                    view.copy_(p.grad)  # preserve grads on mid-run adoption
-                p.grad = view
+                b.offsets.append(off)
+                if self.mode == "alias":
+                    view = _alias_view(b.flat[off:off + p.numel()], p)
+                    if p.grad is not None:
+                        view.copy_(p.grad)  # keep grads on adoption
+                    p.grad = view
                 off = (off + p.numel() + _ALIGN - 1) // _ALIGN * _ALIGN
                 self.bucket_of[p] = b
+            b.plan = None  # pack-mode chunk table, built on first use
 
     def _register_hooks(self):
         for p in self.params:
@@ -153,6 +175,27 @@ class GradBucketReducer:
             self._hooks.append(h)
 
     # -- hot path --------------------------------------------------------
+
+    def _pack_bucket(self, b):
+        """Gather autograd's grad tensors into the flat bucket with one
+        chunk-table kernel launch (pack mode)."""
+        from kungfu_amd.ops import hip as hip_ops
+
+        grads = [p.grad for p in b.params]
+        if any(g is None for g in grads):
+            # rare (a param got no grad): zero the flat, pack the rest
+            b.flat.zero_()
+            for p, off in zip(b.params, b.offsets):
+                if p.grad is not None:
+                    b.flat[off:off + p.numel()].copy_(
+                        p.grad.reshape(-1))
+            return
+        if b.plan is None:
+            b.plan = hip_ops.FusionPlan(grads, b.offsets,
+                                        b.params[0].dtype)
+        else:
+            b.plan.set_sources(grads)
+        b.plan.pack(b.flat)
 
     def _on_grad_ready(self, p):
         b = self.bucket_of[p]
@@ -190,13 +233,23 @@ class GradBucketReducer:
             b = self.buckets[self._launch_seq[self._next_launch]]
             if b.ready < len(b.params):
                 return
-            b.work = self._launch_bucket(b)
+            if self.mode == "pack":
+                self._pack_bucket(b)
+            if self.world > 1:
+                b.work = self._launch_bucket(b)
             b.launched = True
             self._next_launch += 1
 
     def zero_grad(self):
+        if self.mode == "pack":
+            # pack overwrites the flats completely: no zero fill at all;
+            # autograd's next first-accumulation is a steal (grad=None)
+            for p in self.params:
+                p.grad = None
+        else:
+            for b in self.buckets:
+                b.flat.zero_()
         for b in self.buckets:
-            b.flat.zero_()
             b.ready = 0
             b.work = None
             b.launched = False
@@ -206,6 +259,16 @@ class GradBucketReducer:
         """Complete all bucket reductions; call between backward() and
         optimizer.step(). Applies gradient averaging."""
         if self.world <= 1:
+            if self.mode == "pack":
+                # single-rank: no comm, but the fused optimizer reads the
+                # flats — pack any bucket the hooks did not finish
+                if self.overlap and self.is_cuda:
+                    self._drain()
+                for i in self._launch_seq:
+                    b = self.buckets[i]
+                    if not b.launched:
+                        self._pack_bucket(b)
+                        b.launched = True
             return
         if self.is_cuda:
             if self.overlap:
