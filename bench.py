@@ -34,7 +34,7 @@ def parse_args():
     p.add_argument("--optimizer", default="sync",
                    choices=["sync", "sma", "pair", "gns", "sma-gns"])
     p.add_argument("--seq-len", type=int, default=128)
-    p.add_argument("--dtype", default="bf16",
+    p.add_argument("--dtype", default="bf16-master",
                    choices=["bf16", "bf16-master", "bf16-pure", "fp32"],
                    help="bf16 = autocast (fp32 weights, bf16 compute); "
                         "bf16-master = bf16 conv/linear weights with f32 "
