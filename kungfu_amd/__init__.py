@@ -100,7 +100,21 @@ def _init_gpu_backend():
     _set_cuda_device()
     from kungfu_amd.ops import rccl as rccl_ops
 
-    rccl_ops.init_gpu()
+    try:
+        rccl_ops.init_gpu()
+    except Exception as e:
+        # Insurance for unattended multi-GPU runs: a failed native
+        # bootstrap (driver box quirk) must not kill the job silently —
+        # log LOUDLY and fall back to the proven torch.distributed path.
+        import sys
+
+        print("[kungfu] native RCCL bootstrap FAILED (%s); falling back "
+              "to torch.distributed" % e, file=sys.stderr, flush=True)
+        try:
+            rccl_ops.finalize()
+        except Exception:
+            pass
+        _maybe_init_torch_dist()
 
 
 def _maybe_init_torch_dist():
