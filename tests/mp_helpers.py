@@ -900,3 +900,40 @@ def rccl_cpu_elastic_reinit_body(rank, np):
     rccl.finalize()
     kf.finalize()
     return "ok"
+
+
+def rccl_gpu_storm_body(rank, np):
+    """Handle-registry + dispatcher stress: many concurrent async
+    collectives (mixed sizes/types), out-of-order waits, interleaved
+    stream and host completion."""
+    import random
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    torch.cuda.set_device(0)
+    rccl.init_gpu(0)
+    rng = random.Random(1234)
+    tensors = [torch.full((rng.randrange(1, 300_000),), 1.0,
+                          device="cuda") for _ in range(48)]
+    handles = []
+    for i, t in enumerate(tensors):
+        if i % 3 == 0:
+            handles.append(("ar", t, rccl.all_reduce_async(t)))
+        elif i % 3 == 1:
+            handles.append(("bc", t, rccl.broadcast_async(t, root=0)))
+        else:
+            handles.append(("rd", t, rccl.reduce_async(t, root=0)))
+    rng.shuffle(handles)
+    for j, (kind, t, h) in enumerate(handles):
+        if j % 2 == 0:
+            rccl.wait_host(h)
+        else:
+            rccl.wait(h)
+    torch.cuda.synchronize()
+    for kind, t, h in handles:
+        assert float(t[0]) == float(np), (kind, float(t[0]))
+    rccl.finalize()
+    kf.finalize()
+    return True

@@ -30,3 +30,11 @@ def test_rccl_native_two_ranks_one_device(port_block):
                         timeout=300)
     assert all(r == "ok" or str(r).startswith("unsupported") for r in res), res
     print("two-ranks-one-device:", res)
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_rccl_async_storm(port_block):
+    from mp_helpers import rccl_gpu_storm_body
+
+    res = spawn_cluster(rccl_gpu_storm_body, 1, port_block, timeout=300)
+    assert res == [True]
