@@ -27,6 +27,8 @@ hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, long long, int,
                             void *, void *);
 hipError_t kf_bn_fold(void *, int, void *, void *);
+hipError_t kf_col_sum(const void *, long long, int, void *, void *,
+                      void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
                         const void *, long long, int, void *, void *,
@@ -317,6 +319,13 @@ PYBIND11_MODULE(_hip, m)
               check(kf_bn_fold((void *)sums, C, (void *)dbdw,
                                (void *)stream),
                     "kf_bn_fold");
+          });
+    m.def("col_sum",
+          [](uintptr_t dy, long long M, int C, uintptr_t shadows,
+             uintptr_t db, uintptr_t stream) {
+              check(kf_col_sum((const void *)dy, M, C, (void *)shadows,
+                               (void *)db, (void *)stream),
+                    "kf_col_sum");
           });
     m.def("bn_bwd_dx",
           [](uintptr_t dy, uintptr_t x, uintptr_t mask, uintptr_t a,

@@ -9,6 +9,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from kungfu_amd.ops.fused_linear import KfLinear
+
 
 def _make_ln(hidden, fused):
     if fused:
@@ -22,11 +24,13 @@ class BertLayer(nn.Module):
     def __init__(self, hidden, heads, ffn, fused_ln=False):
         super().__init__()
         self.heads = heads
-        self.qkv = nn.Linear(hidden, 3 * hidden)
-        self.proj = nn.Linear(hidden, hidden)
+        # KfLinear = nn.Linear + fused column-sum bias-grad kernel
+        # (native autograd fallback outside the bf16-master conditions)
+        self.qkv = KfLinear(hidden, 3 * hidden)
+        self.proj = KfLinear(hidden, hidden)
         self.ln1 = _make_ln(hidden, fused_ln)
-        self.fc1 = nn.Linear(hidden, ffn)
-        self.fc2 = nn.Linear(ffn, hidden)
+        self.fc1 = KfLinear(hidden, ffn)
+        self.fc2 = KfLinear(ffn, hidden)
         self.ln2 = _make_ln(hidden, fused_ln)
 
     def forward(self, x):

@@ -1,0 +1,73 @@
+"""Linear layer with a fused column-sum bias-gradient kernel.
+
+torch's Linear backward computes grad_bias with the generic column-reduce
+kernel (`at::native::reduce_kernel`), which runs the (b*s, C) bf16 shape
+at ~0.34 TB/s on gfx950 — BERT-base spends ~0.9 ms/step in it. The
+`kf_col_sum` kernel (csrc/hip/fused_bn.hip) uses the shadow-atomic
+per-channel reduction pattern of the fused BN stats pass (~2 TB/s on the
+same shape). Weight/input grads stay on torch matmuls (Tensile MFMA),
+which is exactly what the native backward does.
+
+Fused path conditions: CUDA, bf16 x/weight/bias (the bf16-master training
+mode), out_features % 8 == 0. Anything else falls back to the native
+nn.Linear forward/backward.
+"""
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+try:
+    from kungfu_amd import _hip
+except ImportError:  # pragma: no cover
+    _hip = None
+
+
+class _LinearColSumBias(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, shadows):
+        ctx.save_for_backward(x, weight)
+        ctx.shadows = shadows
+        return F.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        C = weight.shape[0]
+        dy2 = dy.contiguous().reshape(-1, C)
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = (dy2 @ weight).view_as(x)
+        dw = dy2.t() @ x2
+        db = torch.empty(C, dtype=torch.bfloat16, device=dy.device)
+        _hip.col_sum(dy2.data_ptr(), dy2.shape[0], C,
+                     ctx.shadows.data_ptr(), db.data_ptr(),
+                     torch.cuda.current_stream().cuda_stream)
+        return dx, dw, db, None
+
+
+class KfLinear(nn.Linear):
+    """Drop-in nn.Linear; uses the fused bias-grad backward when the
+    bf16-master conditions hold, native autograd otherwise."""
+
+    def __init__(self, *a, **k):
+        super().__init__(*a, **k)
+        self._shadows = None
+
+    def _fused_ok(self, x):
+        return (_hip is not None and x.is_cuda and
+                x.dtype == torch.bfloat16 and
+                self.weight.dtype == torch.bfloat16 and
+                self.bias is not None and
+                self.bias.dtype == torch.bfloat16 and
+                self.out_features % 8 == 0)
+
+    def forward(self, x):
+        if not self._fused_ok(x):
+            return super().forward(x)
+        if self._shadows is None or self._shadows.device != x.device:
+            # NSHADOW=8 interleaved accumulator copies (zeroed once; the
+            # fold kernel re-zeroes after each use)
+            self._shadows = torch.zeros(8 * self.out_features,
+                                        dtype=torch.float32,
+                                        device=x.device)
+        return _LinearColSumBias.apply(x, self.weight, self.bias,
+                                       self._shadows)
