@@ -39,3 +39,39 @@ def elastic_loader(dataset, batch_size, seed=0, **kw):
     loader = torch.utils.data.DataLoader(dataset, batch_size=batch_size,
                                          sampler=sampler, **kw)
     return loader, sampler
+
+
+def mnist_idx(data_dir, train=True):
+    """Load real MNIST from local IDX files (reference
+    kungfu/tensorflow/v1/helpers/mnist.py reads the same format after
+    download; this environment has no network, so files must already be
+    on disk — use synthetic_mnist otherwise). Accepts gzipped or raw
+    idx files with the standard names."""
+    import gzip
+    import os
+    import struct
+
+    import numpy as np
+
+    def read_idx(name):
+        for n in (name, name + ".gz"):
+            p = os.path.join(data_dir, n)
+            if os.path.exists(p):
+                op = gzip.open if n.endswith(".gz") else open
+                with op(p, "rb") as f:
+                    data = f.read()
+                magic, = struct.unpack(">I", data[:4])
+                ndim = magic & 0xff
+                dims = struct.unpack(">" + "I" * ndim,
+                                     data[4:4 + 4 * ndim])
+                arr = np.frombuffer(data, dtype=np.uint8,
+                                    offset=4 + 4 * ndim)
+                return arr.reshape(dims)
+        raise FileNotFoundError("%s(.gz) not in %s" % (name, data_dir))
+
+    prefix = "train" if train else "t10k"
+    x = read_idx("%s-images-idx3-ubyte" % prefix)
+    y = read_idx("%s-labels-idx1-ubyte" % prefix)
+    xt = torch.from_numpy(x.copy()).float().div_(255.0).unsqueeze(1)
+    yt = torch.from_numpy(y.copy()).long()
+    return torch.utils.data.TensorDataset(xt, yt)

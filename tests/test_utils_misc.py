@@ -147,3 +147,27 @@ def test_ingress_accounting(port_block):
 
     res = spawn_cluster(ingress_bytes_body, 2, port_block)
     assert res == [True, True]
+
+
+def test_mnist_idx_roundtrip(tmp_path):
+    """Real-file MNIST loader reads standard IDX (written here, since the
+    environment has no network)."""
+    import gzip
+    import struct
+
+    import numpy as np
+    from kungfu_amd.datasets import mnist_idx
+
+    imgs = np.random.randint(0, 255, (7, 28, 28), dtype=np.uint8)
+    labels = np.random.randint(0, 10, (7,), dtype=np.uint8)
+    with gzip.open(tmp_path / "train-images-idx3-ubyte.gz", "wb") as f:
+        f.write(struct.pack(">I", 0x0803) +
+                struct.pack(">III", 7, 28, 28) + imgs.tobytes())
+    with open(tmp_path / "train-labels-idx1-ubyte", "wb") as f:
+        f.write(struct.pack(">I", 0x0801) + struct.pack(">I", 7) +
+                labels.tobytes())
+    ds = mnist_idx(str(tmp_path))
+    assert len(ds) == 7
+    x0, y0 = ds[0]
+    assert x0.shape == (1, 28, 28) and 0 <= int(y0) < 10
+    assert abs(float(x0[0, 0, 0]) - imgs[0, 0, 0] / 255.0) < 1e-6
