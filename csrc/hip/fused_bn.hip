@@ -521,9 +521,12 @@ __global__ void bn_bwd_dx_kernel(
 // runs this shape at ~0.34 TB/s; this kernel matches bn_stats (~2 TB/s).
 template <int ILP>
 __global__ void col_sum_kernel(const unsigned short *__restrict__ dy,
-                               long long M, int C,
+                               long long M, int C, int ld,
                                float *__restrict__ shadows)
 {
+    // C = aligned column count (multiple of 8); ld = row stride in
+    // elements (>= C; callers with C % 8 != 0 pass the aligned prefix
+    // here and reduce the tail columns separately)
     const int oct_total = C / 8;
     const int oct_base = blockIdx.y * BLOCK;
     const int gpr = min(oct_total - oct_base, BLOCK);
@@ -543,7 +546,7 @@ __global__ void col_sum_kernel(const unsigned short *__restrict__ dy,
             ushort8 v[ILP];
 #pragma unroll
             for (int j = 0; j < ILP; ++j)
-                v[j] = *(const ushort8 *)(dy + (r + j * row_step) * C +
+                v[j] = *(const ushort8 *)(dy + (r + j * row_step) * ld +
                                           col);
 #pragma unroll
             for (int j = 0; j < ILP; ++j) {
@@ -552,7 +555,7 @@ __global__ void col_sum_kernel(const unsigned short *__restrict__ dy,
             }
         }
         for (; r < M; r += row_step) {
-            const ushort8 v = *(const ushort8 *)(dy + r * C + col);
+            const ushort8 v = *(const ushort8 *)(dy + r * ld + col);
 #pragma unroll
             for (int k = 0; k < 8; ++k) s[k] += b2f(v[k]);
         }
@@ -723,10 +726,10 @@ hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *mask,
     return hipGetLastError();
 }
 
-hipError_t kf_col_sum(const void *dy, long long M, int C, void *shadows,
-                      void *db_bf16, void *stream)
+hipError_t kf_col_sum(const void *dy, long long M, int C, int ld,
+                      void *shadows, void *db_bf16, void *stream)
 {
-    if (C % 8 != 0) return hipErrorInvalidValue;
+    if (C % 8 != 0 || ld < C) return hipErrorInvalidValue;
     const int oct_total = C / 8;
     const int gpr = oct_total < BLOCK ? oct_total : BLOCK;
     const int rows_per_blk = BLOCK / gpr;
@@ -737,7 +740,8 @@ hipError_t kf_col_sum(const void *dy, long long M, int C, void *shadows,
     const size_t lds = (size_t)gpr * 8 * sizeof(float);
     const auto s = (hipStream_t)stream;
     hipLaunchKernelGGL((col_sum_kernel<4>), grid, block, lds, s,
-                       (const unsigned short *)dy, M, C, (float *)shadows);
+                       (const unsigned short *)dy, M, C, ld,
+                       (float *)shadows);
     hipLaunchKernelGGL(col_fold_kernel, dim3((C + 255) / 256), dim3(256),
                        0, s, (float *)shadows, C,
                        (unsigned short *)db_bf16);

@@ -27,7 +27,7 @@ hipError_t kf_bn_bwd_reduce(const void *, const void *, const void *,
                             const void *, const void *, long long, int,
                             void *, void *);
 hipError_t kf_bn_fold(void *, int, void *, void *);
-hipError_t kf_col_sum(const void *, long long, int, void *, void *,
+hipError_t kf_col_sum(const void *, long long, int, int, void *, void *,
                       void *);
 hipError_t kf_bn_bwd_dx(const void *, const void *, const void *,
                         const void *, const void *, const void *,
@@ -321,10 +321,11 @@ PYBIND11_MODULE(_hip, m)
                     "kf_bn_fold");
           });
     m.def("col_sum",
-          [](uintptr_t dy, long long M, int C, uintptr_t shadows,
+          [](uintptr_t dy, long long M, int C, int ld, uintptr_t shadows,
              uintptr_t db, uintptr_t stream) {
-              check(kf_col_sum((const void *)dy, M, C, (void *)shadows,
-                               (void *)db, (void *)stream),
+              check(kf_col_sum((const void *)dy, M, C, ld,
+                               (void *)shadows, (void *)db,
+                               (void *)stream),
                     "kf_col_sum");
           });
     m.def("bn_bwd_dx",

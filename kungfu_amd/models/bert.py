@@ -55,7 +55,7 @@ class Bert(nn.Module):
         self.blocks = nn.ModuleList(
             [BertLayer(hidden, heads, ffn, fused_ln=fused_ln)
              for _ in range(layers)])
-        self.head = nn.Linear(hidden, vocab)
+        self.head = KfLinear(hidden, vocab)
         for m in self.modules():
             if isinstance(m, nn.Linear):
                 nn.init.normal_(m.weight, std=0.02)

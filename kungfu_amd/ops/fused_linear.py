@@ -38,9 +38,12 @@ class _LinearColSumBias(torch.autograd.Function):
         dx = (dy2 @ weight).view_as(x)
         dw = dy2.t() @ x2
         db = torch.empty(C, dtype=torch.bfloat16, device=dy.device)
-        _hip.col_sum(dy2.data_ptr(), dy2.shape[0], C,
+        C8 = C & ~7  # kernel covers the 8-aligned prefix ...
+        _hip.col_sum(dy2.data_ptr(), dy2.shape[0], C8, C,
                      ctx.shadows.data_ptr(), db.data_ptr(),
                      torch.cuda.current_stream().cuda_stream)
+        if C8 < C:  # ... tiny unaligned tail via torch (at most 7 cols)
+            db[C8:] = dy2[:, C8:].float().sum(0).to(torch.bfloat16)
         return dx, dw, db, None
 
 
@@ -58,7 +61,7 @@ class KfLinear(nn.Linear):
                 self.weight.dtype == torch.bfloat16 and
                 self.bias is not None and
                 self.bias.dtype == torch.bfloat16 and
-                self.out_features % 8 == 0)
+                self.out_features >= 8)
 
     def forward(self, x):
         if not self._fused_ok(x):

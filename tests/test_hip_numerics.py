@@ -303,7 +303,7 @@ def test_fused_linear_bias_grad():
 
     torch.manual_seed(3)
     for M, C_in, C_out in [(4096, 768, 3072), (4096, 768, 2304),
-                           (127, 64, 8)]:
+                           (127, 64, 8), (256, 32, 30), (64, 16, 30522)]:
         m = KfLinear(C_in, C_out).to("cuda", torch.bfloat16)
         x = torch.randn(M, C_in, device="cuda", dtype=torch.bfloat16,
                         requires_grad=True)
