@@ -177,6 +177,26 @@ PYBIND11_MODULE(_rccl, m)
                                           peer, strm(stream));
           });
 
+    // inline variants: launched on the caller's stream from the calling
+    // thread — ONLY valid under hipGraph stream capture (program order =
+    // agreement; no handles, completion is stream-ordered)
+    m.def("all_reduce_inline",
+          [](int s, uintptr_t send, uintptr_t recv, size_t count,
+             int dtype, int op, uintptr_t stream) {
+              py::gil_scoped_release rel;
+              g.scope(s).all_reduce_inline((const void *)send,
+                                           (void *)recv, count, dt(dtype),
+                                           rop(op), strm(stream));
+          });
+    m.def("broadcast_inline",
+          [](int s, uintptr_t send, uintptr_t recv, size_t count,
+             int dtype, int root, uintptr_t stream) {
+              py::gil_scoped_release rel;
+              g.scope(s).broadcast_inline((const void *)send,
+                                          (void *)recv, count, dt(dtype),
+                                          root, strm(stream));
+          });
+
     // ---- completion ----
     m.def("wait", [](uint64_t h, uintptr_t stream) {
         py::gil_scoped_release rel;

@@ -324,6 +324,24 @@ uint64_t Controller::send_recv(const std::string &name, const void *send,
     });
 }
 
+void Controller::all_reduce_inline(const void *send, void *recv,
+                                   size_t count, DType dt, ReduceOp op,
+                                   hipStream_t stream)
+{
+    rccl_check(ncclAllReduce(send, recv, count, to_rccl_dtype(dt),
+                             to_rccl_op(op), comm().comm(), stream),
+               "ncclAllReduce(inline)");
+}
+
+void Controller::broadcast_inline(const void *send, void *recv,
+                                  size_t count, DType dt, int root,
+                                  hipStream_t stream)
+{
+    rccl_check(ncclBroadcast(send, recv, count, to_rccl_dtype(dt), root,
+                             comm().comm(), stream),
+               "ncclBroadcast(inline)");
+}
+
 void Controller::scheduler_reset(const std::vector<std::string> &names)
 {
     round_names_ = names;

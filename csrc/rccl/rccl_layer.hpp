@@ -110,6 +110,15 @@ class Controller {
                        void *recv, size_t count, DType dt, int peer,
                        hipStream_t caller);
 
+    // Inline launch on the CALLER's stream from the calling thread — for
+    // hipGraph stream capture ONLY (capture is single-threaded, so
+    // program order IS the cross-rank agreement; the dispatcher's
+    // cross-thread events cannot be captured).
+    void all_reduce_inline(const void *send, void *recv, size_t count,
+                           DType dt, ReduceOp op, hipStream_t stream);
+    void broadcast_inline(const void *send, void *recv, size_t count,
+                          DType dt, int root, hipStream_t stream);
+
     // Ordered-release round management (reference scheduler semantics).
     void scheduler_reset(const std::vector<std::string> &names);
     // Broadcast rank-0's last arrival order over the control plane and

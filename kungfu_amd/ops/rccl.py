@@ -108,6 +108,15 @@ def _check(t):
 def all_reduce_async(tensor, op="sum", name="", scope=GLOBAL):
     _check(tensor)
     _ensure_scope(scope)
+    if torch.cuda.is_current_stream_capturing():
+        # hipGraph capture: launch inline on the captured stream (the
+        # dispatcher's cross-thread events cannot be captured; capture is
+        # single-threaded so program order IS the cross-rank agreement)
+        _rccl.all_reduce_inline(scope, tensor.data_ptr(),
+                                tensor.data_ptr(), tensor.numel(),
+                                core_dtype(tensor.dtype), core_op(op),
+                                _stream())
+        return 0  # stream-ordered; wait()/wait_host() are no-ops
     return _rccl.all_reduce(scope, name, tensor.data_ptr(),
                             tensor.data_ptr(), tensor.numel(),
                             core_dtype(tensor.dtype), core_op(op),
@@ -161,10 +170,14 @@ def send_recv_async(send, recv, peer, name="", scope=GLOBAL):
 
 def wait(handle, stream=None):
     """Order the (current) stream after the collective — no host block."""
+    if handle == 0:
+        return  # inline (captured) op: already stream-ordered
     _rccl.wait(handle, stream if stream is not None else _stream())
 
 
 def wait_host(handle):
+    if handle == 0:
+        return
     _rccl.wait_host(handle)
 
 

@@ -937,3 +937,39 @@ def rccl_gpu_storm_body(rank, np):
     rccl.finalize()
     kf.finalize()
     return True
+
+
+def rccl_gpu_graph_capture_body(rank, np):
+    """hipGraph capture of native RCCL collectives via the inline path
+    (capture-safe: launched on the captured stream in program order)."""
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    torch.cuda.set_device(0)
+    rccl.init_gpu(0)
+    x = torch.ones(1 << 20, device="cuda")
+    # warmup on a side stream (graph-capture protocol)
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(3):
+            rccl.wait(rccl.all_reduce_async(x))
+    torch.cuda.current_stream().wait_stream(side)
+    torch.cuda.synchronize()
+    x.fill_(1.0)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        h = rccl.all_reduce_async(x)  # takes the inline path
+        rccl.wait(h)
+        x.mul_(2.0)
+    assert h == 0  # captured ops return the no-op handle
+    for i in range(3):
+        g.replay()
+    torch.cuda.synchronize()
+    # 3 replays of (allreduce(np=1: identity) then *2) from captured state
+    assert float(x[0]) == 2.0 ** 3 * float(np), float(x[0])
+    rccl.finalize()
+    kf.finalize()
+    return True

@@ -38,3 +38,12 @@ def test_rccl_async_storm(port_block):
 
     res = spawn_cluster(rccl_gpu_storm_body, 1, port_block, timeout=300)
     assert res == [True]
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_rccl_graph_capture_inline(port_block):
+    from mp_helpers import rccl_gpu_graph_capture_body
+
+    res = spawn_cluster(rccl_gpu_graph_capture_body, 1, port_block,
+                        timeout=300)
+    assert res == [True]
