@@ -435,6 +435,14 @@ PYBIND11_MODULE(_core, m)
     });
 
     // ---- adaptation / monitoring ----
+    m.def("all_reduce_with",
+          [](const std::vector<int> &parent, uintptr_t sp, uintptr_t r,
+             size_t count, int dt, int op, const std::string &name) {
+              auto w = make_ws(sp, r, count, dt, op, name);
+              py::gil_scoped_release rel;
+              StallGuard sg("all_reduce_with");
+              peer().session().all_reduce_with(parent, w);
+          });
     m.def("set_tree", [](const std::vector<int> &parent) {
         py::gil_scoped_release rel;
         peer().session().set_tree(parent);

@@ -45,6 +45,21 @@ void Session::set_tree(const std::vector<int> &parent)
     stats_.assign(1, {});
 }
 
+void Session::all_reduce_with(const std::vector<int> &parent,
+                              const Workspace &w)
+{
+    if ((int)parent.size() != peers_.size())
+        throw std::runtime_error("all_reduce_with: bad forest size");
+    auto it = forest_cache_.find(parent);
+    if (it == forest_cache_.end()) {
+        it = forest_cache_
+                 .emplace(parent,
+                          std::vector<GraphPair>{gen_from_forest(parent)})
+                 .first;
+    }
+    run_strategies(w, it->second, true);
+}
+
 void Session::set_strategy(Strategy s)
 {
     strategy_kind_ = s;

@@ -67,6 +67,11 @@ class Session {
     // adaptive topology: install a forest (parent array) as the only global
     // strategy (reference adapt.go:45-52 SetTree)
     void set_tree(const std::vector<int> &parent);
+    // one all-reduce over a caller-supplied forest (reference
+    // allreduce.go:18-35 AllReduceWith); the built graph pair is cached
+    // keyed by the parent array
+    void all_reduce_with(const std::vector<int> &parent,
+                         const Workspace &w);
     void set_strategy(Strategy s);
     Strategy strategy() const { return strategy_kind_; }
 
@@ -96,6 +101,7 @@ class Session {
     std::vector<GraphPair> global_, local_, cross_;
     std::vector<GraphPair> reduce_only_;      // built lazily from global_
     std::map<int, GraphPair> root_bcast_;     // cached non-zero-root stars
+    std::map<std::vector<int>, std::vector<GraphPair>> forest_cache_;
     mutable std::mutex stats_mu_;
     std::vector<StrategyStat> stats_;
     double best_throughput_ = 0;

@@ -300,6 +300,30 @@ def hierarchical_all_reduce(tensor, name=None):
     return tensor
 
 
+def monitored_all_reduce(tensor, tree=None, op="sum", name=None):
+    """All-reduce with per-strategy throughput monitoring and an optional
+    per-call spanning tree (reference ops/collective.py:27-44
+    monitored_all_reduce / session AllReduceWith; passing a tree per call
+    is deprecated upstream in favor of set_tree, both are supported).
+    CPU engine only — on CUDA tensors without a tree this is the plain
+    (monitored) all_reduce."""
+    _ensure_init()
+    if _core.size() == 1:
+        return tensor
+    if tree is None:
+        return all_reduce(tensor, op=op, name=name)
+    if tensor.is_cuda:
+        raise ValueError("per-call trees run on the CPU engine; use "
+                         "set_tree for the GPU path")
+    t = tensor.contiguous()
+    _core.all_reduce_with([int(p) for p in tree], t.data_ptr(),
+                          t.data_ptr(), t.numel(), core_dtype(t.dtype),
+                          core_op(op), name or _auto_name("mar", t.numel()))
+    if t.data_ptr() != tensor.data_ptr():
+        tensor.copy_(t)
+    return tensor
+
+
 # ---- model-level helpers ----
 
 def broadcast_parameters(params, root=0):

@@ -973,3 +973,25 @@ def rccl_gpu_graph_capture_body(rank, np):
     rccl.finalize()
     kf.finalize()
     return True
+
+
+def monitored_all_reduce_body(rank, np):
+    """Per-call forest all-reduce (reference AllReduceWith /
+    monitored_all_reduce with a tree): a star rooted at the LAST rank."""
+    import torch
+    import kungfu_amd as kf
+    from kungfu_amd.ops import monitored_all_reduce
+
+    kf.init(with_torch=False)
+    root = np - 1
+    tree = [root] * np  # parent[i] = root; parent[root] = root
+    t = torch.full((4321,), float(rank + 1))
+    monitored_all_reduce(t, tree=tree)
+    expect = float(sum(range(1, np + 1)))
+    assert float(t[0]) == expect and float(t[-1]) == expect
+    # no-tree path = plain monitored all_reduce
+    t2 = torch.full((17,), 2.0)
+    monitored_all_reduce(t2)
+    assert float(t2[0]) == 2.0 * np
+    kf.finalize()
+    return True

@@ -110,3 +110,10 @@ def test_shm_collective_path(strategy, port_block):
     for r in results:
         assert r["small_sum"] == pytest.approx(expect_sum)
         assert r["gathered"] == list(range(4))
+
+
+def test_monitored_all_reduce_with_tree(port_block):
+    from mp_helpers import monitored_all_reduce_body, spawn_cluster
+
+    res = spawn_cluster(monitored_all_reduce_body, 3, port_block)
+    assert res == [True] * 3
