@@ -2,8 +2,10 @@
 
 Reference parity: srcs/python/kungfu/tensorflow/ops/ and kungfu/torch/ops/.
 Routing (MI355X-native):
-  * CUDA tensors -> RCCL over xGMI via torch.distributed (bucketed by the
-    callers in kungfu_amd.parallel);
+  * CUDA tensors -> the NATIVE RCCL layer over xGMI (kungfu_amd.ops.rccl:
+    communicators bootstrapped over the own control plane, stream-ordered
+    handles); torch.distributed only as the KUNGFU_GPU_BACKEND=torch
+    fallback;
   * CPU tensors  -> the C++ collective engine (graph strategies over
     TCP/Unix sockets), zero-copy via data_ptr;
   * `cpu_staged_*` variants stage GPU tensors through host memory into the
