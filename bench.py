@@ -232,6 +232,11 @@ def main():
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda:%d" % torch.cuda.current_device()
                           if use_cuda else "cpu")
+    if args.model in ("vgg16", "inception-v3") and \
+            args.dtype == "bf16-master":
+        # master-weight conversion is wired for resnet50/bert (fused
+        # BN/LN keep f32 norm params); these models run autocast bf16
+        args.dtype = "bf16"
     amp = args.dtype == "bf16"
     dtype = torch.bfloat16 if args.dtype == "bf16-pure" else torch.float32
     if args.channels_last is None:
