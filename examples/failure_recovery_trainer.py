@@ -24,14 +24,16 @@ def main():
     p.add_argument("--restart", type=int, default=0)
     p.add_argument("--crash-at-epoch", type=int, default=-1)
     p.add_argument("--ckpt", default="/tmp/kungfu_fr_ckpt.pt")
+    p.add_argument("--device", default="cpu")
     args = p.parse_args()
 
     kf.init(with_torch=False)
     torch.manual_seed(0)
-    model = SLP(in_features=16, classes=4)
+    model = SLP(in_features=16, classes=4).to(args.device)
     start_epoch = 0
     if args.restart and os.path.exists(args.ckpt + ".%d" % kf.rank()):
-        state = torch.load(args.ckpt + ".%d" % kf.rank())
+        state = torch.load(args.ckpt + ".%d" % kf.rank(),
+                           map_location=args.device)
         model.load_state_dict(state["model"])
         start_epoch = state["epoch"]
         print("RESTARTED from epoch %d" % start_epoch, flush=True)
@@ -42,8 +44,8 @@ def main():
         real_epoch = start_epoch + epoch
         for _ in range(4):
             monitor_batch_begin()
-            x = torch.randn(8, 1, 4, 4)
-            y = torch.randint(0, 4, (8,))
+            x = torch.randn(8, 1, 4, 4, device=args.device)
+            y = torch.randint(0, 4, (8,), device=args.device)
             opt.zero_grad()
             torch.nn.functional.cross_entropy(model(x), y).backward()
             opt.step()

@@ -91,3 +91,39 @@ kf.finalize()
     done = [ln for ln in plain.splitlines() if "DONE" in ln]
     assert len(done) == 3 and all("size=3 step=6" in ln for ln in done), \
         plain
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_auto_recover_gpu_workers(port_block, tmp_path):
+    """Failure detection + auto-recovery with GPU workers (BASELINE §5.3
+    on hardware): rank 0 dies mid-epoch, the heartbeat monitor detects the
+    stall, the runner restarts with adjusted epochs, and the workers
+    reload their checkpoints onto the GPU."""
+    import os
+    import re
+    import sys
+
+    from mp_helpers import run_launcher_graceful
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({
+        "PYTHONPATH": ROOT + os.pathsep + env.get("PYTHONPATH", ""),
+        "HIP_VISIBLE_DEVICES": "0",
+        "CUDA_VISIBLE_DEVICES": "0",
+    })
+    ckpt = str(tmp_path / "ckpt.pt")
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run",
+         "-np", "2", "-port", str(port_block), "-port-range",
+         str(port_block + 1), "-auto-recover", "3s",
+         "-monitor-port", str(port_block + 60),
+         sys.executable, "examples/failure_recovery_trainer.py",
+         "--n-epochs", "4", "--crash-at-epoch", "2", "--ckpt", ckpt,
+         "--device", "cuda"],
+        ROOT, env, 300)
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
+    assert "CRASHING rank 0 now" in plain, plain + err
+    assert "failure detected" in plain, plain
+    ends = [ln for ln in plain.splitlines() if "TRAIN END" in ln]
+    assert len(ends) == 2 and rc == 0, plain + err
