@@ -46,6 +46,13 @@ def _alias_view(flat_slice, p):
     return flat_slice.view_as(p)
 
 
+class _DoneWork:
+    """Already-completed work (synchronous fallback paths)."""
+
+    def wait(self):
+        return True
+
+
 class _NativeBucketWork:
     """Work-alike over a native RCCL handle (stream-ordered wait)."""
 
@@ -212,9 +219,12 @@ class GradBucketReducer:
             self._drain()
 
     def _launch_bucket(self, b):
-        """Issue the bucket's async all-reduce on the active GPU backend:
+        """Issue the bucket's all-reduce on the best available backend:
         native RCCL (default — stream-ordered handle through the per-scope
-        ordered dispatcher) or torch.distributed (fallback)."""
+        ordered dispatcher), torch.distributed (fallback), or host-staged
+        through the C++ engine (the reference's only torch GPU path,
+        src/torch/ops/cuda/collective.cpp — e.g. several CUDA workers
+        sharing one device, where no RCCL communicator can exist)."""
         from kungfu_amd.ops import _native
 
         nat = _native()
@@ -224,7 +234,13 @@ class GradBucketReducer:
             return _NativeBucketWork(h)
         import torch.distributed as dist
 
-        return dist.all_reduce(b.flat, async_op=True)
+        if dist.is_available() and dist.is_initialized():
+            return dist.all_reduce(b.flat, async_op=True)
+        from kungfu_amd.ops import cpu_staged_all_reduce
+
+        cpu_staged_all_reduce(b.flat,
+                              name="%s/%d" % (self.name, b.index))
+        return _DoneWork()
 
     def _drain(self):
         # launch complete buckets strictly in the precomputed launch
