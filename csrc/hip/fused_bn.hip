@@ -77,6 +77,12 @@ inline int bn_oct()
     return oct >= 2 ? 2 : 1;
 }
 
+inline int bn_ilp_stream()
+{
+    static const int ilp = env_int("KF_BN_ILP_STREAM", 2);
+    return ilp >= 4 ? 4 : 2;
+}
+
 // non-temporal loads for the single-use activation streams (KF_BN_NT=1)
 inline bool bn_nt()
 {
@@ -226,8 +232,8 @@ __global__ void bn_finalize_kernel(float *__restrict__ sums,
 
 // ---- pass 2: y = [relu]( a*x + b [+ res] ) ----
 // Fixed channel-octet per thread: per-channel params live in registers for
-// the whole row loop; each block streams contiguous spans. 2-row ILP.
-template <bool RELU, bool RES>
+// the whole row loop; ILP rows in flight per lane (env KF_BN_ILP_STREAM).
+template <bool RELU, bool RES, int ILP>
 __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
                               const unsigned short *__restrict__ res,
                               unsigned short *__restrict__ y,
@@ -248,40 +254,31 @@ __global__ void bn_fwd_kernel(const unsigned short *__restrict__ x,
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
     long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-    for (; r + row_step < M; r += 2 * row_step) {
-        const long long ba = r * C + (long long)g * 8;
-        const long long bb = (r + row_step) * C + (long long)g * 8;
-        const ushort8 va = *(const ushort8 *)(x + ba);
-        const ushort8 vb = *(const ushort8 *)(x + bb);
-        ushort8 rva, rvb;
-        if (RES) {
-            rva = *(const ushort8 *)(res + ba);
-            rvb = *(const ushort8 *)(res + bb);
-        }
-        ushort8 oa, ob;
-        unsigned char ma = 0, mb2 = 0;
+    for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
+        ushort8 v[ILP], rv[ILP];
+        long long base[ILP];
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-            float ua = fmaf(ar[k], b2f(va[k]), br[k]);
-            float ub = fmaf(ar[k], b2f(vb[k]), br[k]);
-            if (RES) {
-                ua += b2f(rva[k]);
-                ub += b2f(rvb[k]);
-            }
-            if (RELU) {
-                if (ua > 0.f) ma |= (unsigned char)(1u << k);
-                if (ub > 0.f) mb2 |= (unsigned char)(1u << k);
-                ua = fmaxf(ua, 0.f);
-                ub = fmaxf(ub, 0.f);
-            }
-            oa[k] = f2b(ua);
-            ob[k] = f2b(ub);
+        for (int j = 0; j < ILP; ++j) {
+            base[j] = (r + j * row_step) * C + (long long)g * 8;
+            v[j] = *(const ushort8 *)(x + base[j]);
+            if (RES) rv[j] = *(const ushort8 *)(res + base[j]);
         }
-        *(ushort8 *)(y + ba) = oa;
-        *(ushort8 *)(y + bb) = ob;
-        if (RELU && mask) {
-            mask[r * gpr + g] = ma;
-            mask[(r + row_step) * gpr + g] = mb2;
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+            ushort8 out;
+            unsigned char mb = 0;
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                float u = fmaf(ar[k], b2f(v[j][k]), br[k]);
+                if (RES) u += b2f(rv[j][k]);
+                if (RELU) {
+                    if (u > 0.f) mb |= (unsigned char)(1u << k);
+                    u = fmaxf(u, 0.f);
+                }
+                out[k] = f2b(u);
+            }
+            *(ushort8 *)(y + base[j]) = out;
+            if (RELU && mask) mask[(r + j * row_step) * gpr + g] = mb;
         }
     }
     for (; r < M; r += row_step) {
@@ -430,8 +427,8 @@ __global__ void bn_fold_kernel(float *__restrict__ sums, int C,
 
 // ---- backward pass 2: dx (and d_res when fused residual) ----
 // dx = w*rstd * (dy_m - sum_dy/M - xhat * sum_dyxhat/M); d_res = dy_m.
-// 2-row ILP.
-template <bool MASKED, bool RES>
+// ILP rows in flight per lane (env KF_BN_ILP_STREAM).
+template <bool MASKED, bool RES, int ILP>
 __global__ void bn_bwd_dx_kernel(
     const unsigned short *__restrict__ dy,
     const unsigned short *__restrict__ x,
@@ -458,38 +455,31 @@ __global__ void bn_bwd_dx_kernel(
     }
     const long long row_step = (long long)gridDim.x * rows_per_blk;
     long long r = (long long)blockIdx.x * rows_per_blk + row_off;
-    for (; r + row_step < M; r += 2 * row_step) {
-        const long long ba = r * C + (long long)g * 8;
-        const long long bb = (r + row_step) * C + (long long)g * 8;
-        const ushort8 dva = *(const ushort8 *)(dy + ba);
-        const ushort8 xva = *(const ushort8 *)(x + ba);
-        const ushort8 dvb = *(const ushort8 *)(dy + bb);
-        const ushort8 xvb = *(const ushort8 *)(x + bb);
-        const unsigned char ma =
-            MASKED ? mask[r * gpr + g] : (unsigned char)0xff;
-        const unsigned char mbm =
-            MASKED ? mask[(r + row_step) * gpr + g] : (unsigned char)0xff;
-        ushort8 dxa, dxb, dra, drb;
+    for (; r + (ILP - 1) * row_step < M; r += ILP * row_step) {
+        ushort8 dv[ILP], xv[ILP];
+        unsigned char mk[ILP];
+        long long base[ILP];
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-            float da = b2f(dva[k]);
-            float db = b2f(dvb[k]);
-            if (MASKED && !((ma >> k) & 1)) da = 0.f;
-            if (MASKED && !((mbm >> k) & 1)) db = 0.f;
-            if (RES) {
-                dra[k] = f2b(da);
-                drb[k] = f2b(db);
-            }
-            const float xha = (b2f(xva[k]) - mr[k]) * rr[k];
-            const float xhb = (b2f(xvb[k]) - mr[k]) * rr[k];
-            dxa[k] = f2b(ar[k] * (da - t1[k] - xha * t2[k]));
-            dxb[k] = f2b(ar[k] * (db - t1[k] - xhb * t2[k]));
+        for (int j = 0; j < ILP; ++j) {
+            base[j] = (r + j * row_step) * C + (long long)g * 8;
+            dv[j] = *(const ushort8 *)(dy + base[j]);
+            xv[j] = *(const ushort8 *)(x + base[j]);
+            mk[j] = MASKED ? mask[(r + j * row_step) * gpr + g]
+                           : (unsigned char)0xff;
         }
-        *(ushort8 *)(dx + ba) = dxa;
-        *(ushort8 *)(dx + bb) = dxb;
-        if (RES) {
-            *(ushort8 *)(dres + ba) = dra;
-            *(ushort8 *)(dres + bb) = drb;
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+            ushort8 dxo, dro;
+#pragma unroll
+            for (int k = 0; k < 8; ++k) {
+                float d = b2f(dv[j][k]);
+                if (MASKED && !((mk[j] >> k) & 1)) d = 0.f;
+                if (RES) dro[k] = f2b(d);
+                const float xh = (b2f(xv[j][k]) - mr[k]) * rr[k];
+                dxo[k] = f2b(ar[k] * (d - t1[k] - xh * t2[k]));
+            }
+            *(ushort8 *)(dx + base[j]) = dxo;
+            if (RES) *(ushort8 *)(dres + base[j]) = dro;
         }
     }
     for (; r < M; r += row_step) {
@@ -644,16 +634,23 @@ hipError_t kf_bn_fwd(const void *x, const void *res, void *y, const void *a,
     if (blocks < 1) blocks = 1;
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
-#define CASE(R, E)                                                          \
-    hipLaunchKernelGGL((bn_fwd_kernel<R, E>), grid, block, 0, s,            \
+#define CASE(R, E, I)                                                       \
+    hipLaunchKernelGGL((bn_fwd_kernel<R, E, I>), grid, block, 0, s,         \
                        (const unsigned short *)x,                           \
                        (const unsigned short *)res, (unsigned short *)y,    \
                        (const float *)a, (const float *)b, M, C,            \
                        (unsigned char *)mask)
-    if (relu && res) CASE(true, true);
-    else if (relu) CASE(true, false);
-    else if (res) CASE(false, true);
-    else CASE(false, false);
+    if (bn_ilp_stream() == 4) {
+        if (relu && res) CASE(true, true, 4);
+        else if (relu) CASE(true, false, 4);
+        else if (res) CASE(false, true, 4);
+        else CASE(false, false, 4);
+    } else {
+        if (relu && res) CASE(true, true, 2);
+        else if (relu) CASE(true, false, 2);
+        else if (res) CASE(false, true, 2);
+        else CASE(false, false, 2);
+    }
 #undef CASE
     return hipGetLastError();
 }
@@ -710,18 +707,25 @@ hipError_t kf_bn_bwd_dx(const void *dy, const void *x, const void *mask,
     if (blocks < 1) blocks = 1;
     const dim3 grid((uint32_t)blocks), block(BLOCK);
     const auto s = (hipStream_t)stream;
-#define CASE(MK, E)                                                         \
-    hipLaunchKernelGGL((bn_bwd_dx_kernel<MK, E>), grid, block, 0, s,        \
+#define CASE(MK, E, I)                                                      \
+    hipLaunchKernelGGL((bn_bwd_dx_kernel<MK, E, I>), grid, block, 0, s,     \
                        (const unsigned short *)dy,                          \
                        (const unsigned short *)x,                           \
                        (const unsigned char *)mask, (const float *)a,       \
                        (const float *)mean, (const float *)rstd,            \
                        (const float *)sums, M, C, (unsigned short *)dx,     \
                        (unsigned short *)dres)
-    if (mask && dres) CASE(true, true);
-    else if (mask) CASE(true, false);
-    else if (dres) CASE(false, true);
-    else CASE(false, false);
+    if (bn_ilp_stream() == 4) {
+        if (mask && dres) CASE(true, true, 4);
+        else if (mask) CASE(true, false, 4);
+        else if (dres) CASE(false, true, 4);
+        else CASE(false, false, 4);
+    } else {
+        if (mask && dres) CASE(true, true, 2);
+        else if (mask) CASE(true, false, 2);
+        else if (dres) CASE(false, true, 2);
+        else CASE(false, false, 2);
+    }
 #undef CASE
     return hipGetLastError();
 }
