@@ -15,8 +15,15 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 namespace {
+
+inline int env_int(const char *name, int dflt)
+{
+    const char *v = getenv(name);
+    return v && *v ? atoi(v) : dflt;
+}
 
 constexpr int WAVE = 64;
 constexpr int BLOCK = 256;              // 4 waves = 4 rows per block
@@ -247,9 +254,17 @@ hipError_t kf_ln_bwd(const void *dy, const void *x, const void *w,
 {
     if (H % 8 != 0 || H > WAVE * 8 * MAX_K) return hipErrorInvalidValue;
     long long blocks = (N + ROWS_PER_BLOCK - 1) / ROWS_PER_BLOCK;
-    const long long by_bytes = (N * H * 4 + 65535) / 65536;
-    if (blocks > by_bytes) blocks = by_bytes;
-    if (blocks > 2048) blocks = 2048;
+    // KF_LN_CAPKB: min KiB of dy+x input per block (0 = off);
+    // KF_LN_MAXBLK: hard cap. BERT-base shapes (N=4096, H=768) are small
+    // enough that the old 64 KiB floor left the chip <1 block/CU.
+    static const int cap_kb = env_int("KF_LN_CAPKB", 16);
+    static const int maxblk = env_int("KF_LN_MAXBLK", 1024);
+    if (cap_kb > 0) {
+        const long long by_bytes =
+            (N * H * 4 + cap_kb * 1024 - 1) / (cap_kb * 1024);
+        if (blocks > by_bytes) blocks = by_bytes;
+    }
+    if (blocks > maxblk) blocks = maxblk;
     if (blocks < 1) blocks = 1;
     const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
 #define CASE(KK)                                                            \
