@@ -194,3 +194,50 @@ def test_bf16_master_training_converges():
     if b.master is not None:
         assert torch.equal(b.param_flat,
                            b.master.to(torch.bfloat16))
+
+
+def test_checkpoint_roundtrip_fused_master(tmp_path):
+    """Checkpoint/restore through the fused-master optimizer: f32 masters
+    and momentum live in the fused flat buffers, and restored training
+    must continue bit-identically."""
+    import kungfu_amd as kf
+    from kungfu_amd.models import SLP
+    from kungfu_amd.optimizers import SynchronousSGDOptimizer
+    from kungfu_amd.parallel.elastic import (load_checkpoint,
+                                             save_checkpoint)
+    from kungfu_amd.utils.precision import convert_bf16_master
+
+    kf.init()
+    torch.manual_seed(6)
+
+    def make():
+        m = convert_bf16_master(SLP(in_features=8, classes=3).to("cuda"))
+        o = SynchronousSGDOptimizer(
+            torch.optim.SGD(m.parameters(), lr=0.1, momentum=0.9),
+            fused_step=True)
+        return m, o
+
+    model, opt = make()
+    x = torch.randn(4, 1, 2, 4, device="cuda", dtype=torch.bfloat16)
+    y = torch.randint(0, 3, (4,), device="cuda")
+    for _ in range(3):
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(model(x).float(), y).backward()
+        opt.step()
+    torch.cuda.synchronize()
+    path = str(tmp_path / "ck")
+    save_checkpoint(path, model, opt, step=3)
+    model2, opt2 = make()
+    step, _ = load_checkpoint(path, model2, opt2, map_location="cuda")
+    assert step == 3
+    for b1, b2 in zip(opt.reducer.buckets, opt2.reducer.buckets):
+        if b1.master is not None:
+            assert torch.equal(b1.master, b2.master)
+        assert torch.equal(b1.momentum, b2.momentum)
+    for m, o in ((model, opt), (model2, opt2)):
+        o.zero_grad()
+        torch.nn.functional.cross_entropy(m(x).float(), y).backward()
+        o.step()
+    torch.cuda.synchronize()
+    for b1, b2 in zip(opt.reducer.buckets, opt2.reducer.buckets):
+        assert torch.equal(b1.param_flat, b2.param_flat)

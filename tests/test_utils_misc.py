@@ -171,3 +171,44 @@ def test_mnist_idx_roundtrip(tmp_path):
     x0, y0 = ds[0]
     assert x0.shape == (1, 28, 28) and 0 <= int(y0) < 10
     assert abs(float(x0[0, 0, 0]) - imgs[0, 0, 0] / 255.0) < 1e-6
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    """save_checkpoint/load_checkpoint restore model+optimizer+step
+    exactly (single-process; the fused-master GPU variant lives in
+    test_gpu_training)."""
+    import torch
+
+    import kungfu_amd as kf
+    from kungfu_amd.models import SLP
+    from kungfu_amd.optimizers import SynchronousSGDOptimizer
+    from kungfu_amd.parallel.elastic import (load_checkpoint,
+                                             save_checkpoint)
+
+    kf.init(with_torch=False)
+    torch.manual_seed(5)
+    model = SLP(in_features=8, classes=3)
+    opt = SynchronousSGDOptimizer(
+        torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9))
+    x = torch.randn(4, 1, 2, 4)
+    y = torch.randint(0, 3, (4,))
+    for _ in range(3):
+        opt.zero_grad()
+        torch.nn.functional.cross_entropy(model(x), y).backward()
+        opt.step()
+    path = str(tmp_path / "ck")
+    save_checkpoint(path, model, opt, step=3, extra={"lr": 0.1})
+    model2 = SLP(in_features=8, classes=3)
+    opt2 = SynchronousSGDOptimizer(
+        torch.optim.SGD(model2.parameters(), lr=0.1, momentum=0.9))
+    step, extra = load_checkpoint(path, model2, opt2)
+    assert step == 3 and extra == {"lr": 0.1}
+    for p1, p2 in zip(model.parameters(), model2.parameters()):
+        assert torch.equal(p1, p2)
+    # training continues identically from the restored state
+    for m, o in ((model, opt), (model2, opt2)):
+        o.zero_grad()
+        torch.nn.functional.cross_entropy(m(x), y).backward()
+        o.step()
+    for p1, p2 in zip(model.parameters(), model2.parameters()):
+        assert torch.equal(p1, p2)
