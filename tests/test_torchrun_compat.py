@@ -138,3 +138,39 @@ def test_tournament_partner_schedule():
             for r in range(n):
                 met = {tournament_partner(r, s, n) for s in range(n - 1)}
                 assert met == set(range(n)) - {r}, (n, r, met)
+
+
+@retry_flaky
+def test_bench_resnet50_n2_cpu_orchestration(port_block):
+    """The FULL flagship orchestration at N=2 on CPU (driver-style env:
+    torchrun synthesis, model broadcast, bucketed gradient all-reduce
+    through the C++ engine, max-over-ranks timing, single JSON line) —
+    everything the 8-GPU run does except the RCCL transport itself."""
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update({
+            "PYTHONPATH": ROOT,
+            "RANK": str(rank),
+            "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": "2",
+            "MASTER_ADDR": "127.0.0.1",
+            "MASTER_PORT": str(port_block + 41),
+            "KUNGFU_PORT_BASE": str(port_block),
+        })
+        procs.append(subprocess.Popen(
+            [sys.executable, "bench.py", "--model", "resnet50",
+             "--batch-size", "2", "--steps", "2", "--warmup", "1",
+             "--dtype", "fp32", "--fused-bn", "0", "--fused-opt", "0",
+             "--graph", "0"],
+            cwd=ROOT, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        out, err = p.communicate(timeout=300)
+        assert p.returncode == 0, out + err
+        outs.append(out)
+    jsons = [ln for out in outs for ln in out.splitlines()
+             if ln.startswith("{")]
+    assert len(jsons) == 1, outs
+    assert '"n_gpus": 2' in jsons[0] and '"global_batch": 4' in jsons[0]
