@@ -224,6 +224,15 @@ def main():
                           "tuned", "miopen")
     if os.path.isdir(_tuned):
         os.environ.setdefault("MIOPEN_USER_DB_PATH", _tuned)
+    # hipBLASLt GEMM selections pre-tuned via PyTorch TunableOp (BERT
+    # +2.7% same-box); TUNING stays off so untuned shapes just use the
+    # default picks
+    _gemm = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                         "tuned", "tunableop", "gemm.csv")
+    if os.path.exists(os.path.join(os.path.dirname(_gemm), "gemm0.csv")):
+        os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "0")
+        os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _gemm)
     torch.backends.cudnn.benchmark = True  # MIOpen find once per shape
     import kungfu_amd as kf
 
