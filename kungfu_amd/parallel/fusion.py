@@ -292,13 +292,21 @@ class GradBucketReducer:
                 for i in self._launch_seq:
                     b = self.buckets[i]
                     if not b.launched:  # param got no grad this step
+                        if self.mode == "pack":
+                            self._pack_bucket(b)
                         b.work = self._launch_bucket(b)
                         b.launched = True
                 for b in self.buckets:
                     b.work.wait()
             else:
                 for i in self._launch_seq:
-                    self._launch_bucket(self.buckets[i]).wait()
+                    b = self.buckets[i]
+                    if self.mode == "pack":
+                        # non-overlap pack mode (e.g. gradient
+                        # accumulation): gather the accumulated autograd
+                        # grads before reducing
+                        self._pack_bucket(b)
+                    self._launch_bucket(b).wait()
             if self.average:
                 for b in self.buckets:
                     b.flat.div_(self.world)

@@ -241,3 +241,31 @@ def test_checkpoint_roundtrip_fused_master(tmp_path):
     torch.cuda.synchronize()
     for b1, b2 in zip(opt.reducer.buckets, opt2.reducer.buckets):
         assert torch.equal(b1.param_flat, b2.param_flat)
+
+
+def test_pack_mode_gradient_accumulation():
+    """overlap=False + several backwards per step (gradient accumulation)
+    in pack mode: the packed flats must hold the ACCUMULATED autograd
+    grads (regression: finalize once launched without packing)."""
+    import kungfu_amd as kf
+    from kungfu_amd.optimizers import SynchronousSGDOptimizer
+
+    kf.init()
+    torch.manual_seed(9)
+    lin = torch.nn.Linear(32, 16).to("cuda", torch.bfloat16)
+    opt = SynchronousSGDOptimizer(
+        torch.optim.SGD(lin.parameters(), lr=0.0, momentum=0.0),
+        overlap=False, fused_step=True)
+    assert opt.reducer.mode == "pack"
+    opt.zero_grad()
+    xs = [torch.randn(8, 32, device="cuda", dtype=torch.bfloat16)
+          for _ in range(3)]
+    for x in xs:
+        lin(x).sum().backward()
+    accumulated = lin.weight.grad.clone()
+    opt.step()  # lr=0: packs + (world-1) fused apply, params unchanged
+    torch.cuda.synchronize()
+    b = opt.reducer.bucket_of[lin.weight]
+    off = b.offsets[b.params.index(lin.weight)]
+    flat = b.flat[off:off + lin.weight.numel()]
+    assert torch.equal(flat, accumulated.reshape(-1))
