@@ -266,6 +266,7 @@ def test_pack_mode_gradient_accumulation():
     opt.step()  # lr=0: packs + (world-1) fused apply, params unchanged
     torch.cuda.synchronize()
     b = opt.reducer.bucket_of[lin.weight]
-    off = b.offsets[b.params.index(lin.weight)]
+    idx = next(i for i, p in enumerate(b.params) if p is lin.weight)
+    off = b.offsets[idx]
     flat = b.flat[off:off + lin.weight.numel()]
     assert torch.equal(flat, accumulated.reshape(-1))
