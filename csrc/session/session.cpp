@@ -40,7 +40,7 @@ void Session::set_tree(const std::vector<int> &parent)
     if ((int)parent.size() != peers_.size())
         throw std::runtime_error("set_tree: bad forest size");
     global_ = {gen_from_forest(parent)};
-    reduce_only_.clear();
+    reduce_only_.reset();
     std::lock_guard<std::mutex> lk(stats_mu_);
     stats_.assign(1, {});
 }
@@ -50,21 +50,28 @@ void Session::all_reduce_with(const std::vector<int> &parent,
 {
     if ((int)parent.size() != peers_.size())
         throw std::runtime_error("all_reduce_with: bad forest size");
-    auto it = forest_cache_.find(parent);
-    if (it == forest_cache_.end()) {
-        it = forest_cache_
-                 .emplace(parent,
-                          std::vector<GraphPair>{gen_from_forest(parent)})
-                 .first;
+    std::shared_ptr<std::vector<GraphPair>> g;
+    {
+        std::lock_guard<std::mutex> lk(stats_mu_);
+        auto it = forest_cache_.find(parent);
+        if (it == forest_cache_.end()) {
+            it = forest_cache_
+                     .emplace(parent,
+                              std::make_shared<std::vector<GraphPair>>(
+                                  std::vector<GraphPair>{
+                                      gen_from_forest(parent)}))
+                     .first;
+        }
+        g = it->second;
     }
-    run_strategies(w, it->second, true);
+    run_strategies(w, *g, true);
 }
 
 void Session::set_strategy(Strategy s)
 {
     strategy_kind_ = s;
     global_ = gen_strategies(peers_, s);
-    reduce_only_.clear();
+    reduce_only_.reset();
     std::lock_guard<std::mutex> lk(stats_mu_);
     stats_.assign(global_.size(), {});
 }
@@ -235,16 +242,22 @@ void Session::reduce(const Workspace &w)
     // from ALL global strategies so chunked reduces rotate across
     // topologies exactly like all-reduce (round-1 gap: always used
     // global_[0] and ran unchunked)
-    if (reduce_only_.size() != global_.size()) {
-        reduce_only_.clear();
-        for (const auto &g : global_) {
-            GraphPair p;
-            p.reduce = g.reduce;
-            p.bcast = Graph(p.reduce.n);  // result stays at root
-            reduce_only_.push_back(std::move(p));
+    std::shared_ptr<std::vector<GraphPair>> ro;
+    {
+        std::lock_guard<std::mutex> lk(stats_mu_);
+        if (!reduce_only_ || reduce_only_->size() != global_.size()) {
+            auto fresh = std::make_shared<std::vector<GraphPair>>();
+            for (const auto &g : global_) {
+                GraphPair p;
+                p.reduce = g.reduce;
+                p.bcast = Graph(p.reduce.n);  // result stays at root
+                fresh->push_back(std::move(p));
+            }
+            reduce_only_ = std::move(fresh);
         }
+        ro = reduce_only_;
     }
-    run_strategies(w, reduce_only_, false);
+    run_strategies(w, *ro, false);
 }
 
 void Session::broadcast(const Workspace &w, int root)

@@ -15,6 +15,7 @@
 
 #include <functional>
 #include <map>
+#include <memory>
 #include <string>
 #include <vector>
 
@@ -99,9 +100,12 @@ class Session {
     CollectiveEndpoint &collective_;
     Strategy strategy_kind_;
     std::vector<GraphPair> global_, local_, cross_;
-    std::vector<GraphPair> reduce_only_;      // built lazily from global_
+    // lazily built graph caches; guarded by stats_mu_ (async collective
+    // handles may run Session methods concurrently)
+    std::shared_ptr<std::vector<GraphPair>> reduce_only_;
     std::map<int, GraphPair> root_bcast_;     // cached non-zero-root stars
-    std::map<std::vector<int>, std::vector<GraphPair>> forest_cache_;
+    std::map<std::vector<int>,
+             std::shared_ptr<std::vector<GraphPair>>> forest_cache_;
     mutable std::mutex stats_mu_;
     std::vector<StrategyStat> stats_;
     double best_throughput_ = 0;
