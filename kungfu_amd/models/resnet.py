@@ -7,32 +7,7 @@ same architecture from tf.keras.applications
 canonical model: 161 trainable tensors / ~25.6M params
 (tests/go/fakemodel/resnet50-imagenet.go).
 """
-import torch
 import torch.nn as nn
-
-
-class Conv1x1(nn.Conv2d):
-    """1x1 stride-1 convolution executed as a hipBLASLt GEMM on NHWC
-    tensors: for channels_last activations a pointwise conv IS
-    x[M,Cin] @ W[Cin,Cout] with zero-copy views on both sides (the GEMM
-    output in row-major [M,Cout] is exactly the channels_last layout the
-    fused BN consumes). Routes ~36 of ResNet-50's 53 convs through
-    Tensile MFMA kernels that TunableOp can pre-tune (tuned/tunableop)
-    instead of MIOpen/CK. Falls back to the normal conv path for strided
-    or non-NHWC cases (stride-2 downsamples, CPU, autocast dtype mix).
-    For 1x1 kernels the weight's channels_last and row-major layouts
-    coincide, so the view below never copies."""
-
-    def forward(self, x):
-        if (x.is_cuda and x.dim() == 4 and
-                x.dtype == self.weight.dtype and
-                self.stride == (1, 1) and
-                x.is_contiguous(memory_format=torch.channels_last)):
-            n, c, h, w = x.shape
-            x2 = x.permute(0, 2, 3, 1).reshape(-1, c)
-            y2 = x2 @ self.weight.view(self.out_channels, c).t()
-            return y2.view(n, h, w, self.out_channels).permute(0, 3, 1, 2)
-        return super().forward(x)
 
 
 class Bottleneck(nn.Module):
@@ -42,10 +17,10 @@ class Bottleneck(nn.Module):
                  fused_bn=False):
         super().__init__()
         cout = width * self.expansion
-        self.conv1 = Conv1x1(cin, width, 1, bias=False)
+        self.conv1 = nn.Conv2d(cin, width, 1, bias=False)
         self.conv2 = nn.Conv2d(width, width, 3, stride=stride, padding=1,
                                bias=False)
-        self.conv3 = Conv1x1(width, cout, 1, bias=False)
+        self.conv3 = nn.Conv2d(width, cout, 1, bias=False)
         self.fused = fused_bn
         if fused_bn:
             from kungfu_amd.ops.fused_bn import FusedBNReLU2d
@@ -114,9 +89,7 @@ class ResNet(nn.Module):
             else:
                 ds_bn = nn.BatchNorm2d(cout)
             downsample = nn.Sequential(
-                # stride-2 downsamples fall back to the conv path inside
-                # Conv1x1; stride-1 projections take the GEMM route
-                Conv1x1(self.cin, cout, 1, stride=stride, bias=False),
+                nn.Conv2d(self.cin, cout, 1, stride=stride, bias=False),
                 ds_bn,
             )
         layers = [Bottleneck(self.cin, width, stride, downsample,

@@ -325,37 +325,3 @@ def test_fused_linear_bias_grad():
         ref_db2 = dy.float().sum(0)
         assert torch.allclose(m.bias.grad.float(), ref_db2, rtol=2e-2,
                               atol=2e-1)
-
-
-def test_conv1x1_gemm_matches_conv():
-    """Conv1x1's GEMM route must equal the conv route bit-for-bit in f32
-    and within bf16 tolerance, fwd and bwd, on channels_last tensors."""
-    from kungfu_amd.models.resnet import Conv1x1
-
-    torch.manual_seed(11)
-    for dtype, tol in ((torch.float32, 1e-4), (torch.bfloat16, 2e-2)):
-        m = Conv1x1(64, 128, 1, bias=False).to("cuda", dtype)
-        ref = torch.nn.Conv2d(64, 128, 1, bias=False).to("cuda", dtype)
-        ref.load_state_dict(m.state_dict())
-        x1 = torch.randn(4, 64, 14, 14, device="cuda", dtype=dtype,
-                         requires_grad=True)
-        x1g = x1.contiguous(memory_format=torch.channels_last)
-        x2 = x1.detach().clone().requires_grad_()
-        y1 = m(x1g)
-        assert y1.is_contiguous(memory_format=torch.channels_last)
-        y2 = ref(x2)
-        assert torch.allclose(y1.float(), y2.float(), atol=tol, rtol=tol)
-        g = torch.randn_like(y1)
-        y1.backward(g)
-        y2.backward(g.contiguous())
-        torch.cuda.synchronize()
-        assert torch.allclose(m.weight.grad.float(),
-                              ref.weight.grad.float(), atol=tol * 10,
-                              rtol=tol * 10)
-        assert torch.allclose(x1.grad.float(), x2.grad.float(),
-                              atol=tol * 10, rtol=tol * 10)
-    # stride-2 falls back to the conv path
-    ms = Conv1x1(8, 16, 1, stride=2, bias=False).to("cuda")
-    xs = torch.randn(2, 8, 8, 8, device="cuda").contiguous(
-        memory_format=torch.channels_last)
-    assert ms(xs).shape == (2, 16, 4, 4)
