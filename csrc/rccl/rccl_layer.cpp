@@ -62,6 +62,14 @@ ncclRedOp_t to_rccl_op(ReduceOp op)
 
 // ---- GpuHandle ----
 
+GpuHandle::~GpuHandle()
+{
+    // normal waits destroy the events and null the fields; this covers
+    // error paths where the handle is dropped without a wait
+    if (ready) (void)hipEventDestroy(ready);
+    if (done) (void)hipEventDestroy(done);
+}
+
 void GpuHandle::mark_launched(std::string err)
 {
     {

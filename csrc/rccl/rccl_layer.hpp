@@ -48,6 +48,7 @@ struct GpuHandle {
     bool launched = false;
     std::string error;  // non-empty => launch failed on dispatcher thread
 
+    ~GpuHandle();  // frees any events not released by a wait (error paths)
     void mark_launched(std::string err);
     void wait_launched();  // throws on captured error
 };
