@@ -121,7 +121,9 @@ __global__ void ln_fwd_kernel(const unsigned short *__restrict__ x,
 // ---- backward (single pass) ----
 // dx = rstd * (g - mean(g) - xhat * mean(g * xhat)),  g = dy * w
 // dW[h] += sum_rows dy * xhat ; db[h] += sum_rows dy  (via LDS + shadows)
-template <int K>
+// ILP independent rows in flight per wave: the two serial phases per
+// row (loads -> wave_sum shuffle chains -> store) pipeline across rows.
+template <int K, int ILP>
 __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
                               const unsigned short *__restrict__ x,
                               const float *__restrict__ w,
@@ -143,8 +145,71 @@ __global__ void ln_bwd_kernel(const unsigned short *__restrict__ dy,
     float accw[K][8] = {};
     float accb[K][8] = {};
     const long long row_step = (long long)gridDim.x * ROWS_PER_BLOCK;
-    for (long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid; r < N;
-         r += row_step) {
+    long long r = (long long)blockIdx.x * ROWS_PER_BLOCK + wid;
+    for (; r + (ILP - 1) * row_step < N; r += ILP * row_step) {
+        ushort8 dv[ILP][K], xv[ILP][K];
+        float s1[ILP], s2[ILP], mean[ILP], rstd[ILP];
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+            const long long rj = r + j * row_step;
+            mean[j] = save_mean[rj];
+            rstd[j] = save_rstd[rj];
+            s1[j] = s2[j] = 0.f;
+#pragma unroll
+            for (int k = 0; k < K; ++k) {
+                const int h0 = (k * WAVE + lane) * 8;
+                if (h0 < H) {
+                    dv[j][k] = *(const ushort8 *)(dy + rj * H + h0);
+                    xv[j][k] = *(const ushort8 *)(x + rj * H + h0);
+                }
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+#pragma unroll
+            for (int k = 0; k < K; ++k) {
+                const int h0 = (k * WAVE + lane) * 8;
+                if (h0 < H) {
+#pragma unroll
+                    for (int i = 0; i < 8; ++i) {
+                        const float d = b2f(dv[j][k][i]);
+                        const float xh =
+                            (b2f(xv[j][k][i]) - mean[j]) * rstd[j];
+                        s1[j] += d * w[h0 + i];
+                        s2[j] += d * w[h0 + i] * xh;
+                    }
+                }
+            }
+        }
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+            s1[j] = wave_sum(s1[j]) / (float)H;
+            s2[j] = wave_sum(s2[j]) / (float)H;
+        }
+#pragma unroll
+        for (int j = 0; j < ILP; ++j) {
+            const long long rj = r + j * row_step;
+#pragma unroll
+            for (int k = 0; k < K; ++k) {
+                const int h0 = (k * WAVE + lane) * 8;
+                if (h0 < H) {
+                    ushort8 o;
+#pragma unroll
+                    for (int i = 0; i < 8; ++i) {
+                        const float d = b2f(dv[j][k][i]);
+                        const float xh =
+                            (b2f(xv[j][k][i]) - mean[j]) * rstd[j];
+                        const float g = d * w[h0 + i];
+                        o[i] = f2b(rstd[j] * (g - s1[j] - xh * s2[j]));
+                        accw[k][i] += d * xh;
+                        accb[k][i] += d;
+                    }
+                    *(ushort8 *)(dx + rj * H + h0) = o;
+                }
+            }
+        }
+    }
+    for (; r < N; r += row_step) {
         const unsigned short *dyr = dy + r * H;
         const unsigned short *xr = x + r * H;
         const float mean = save_mean[r];
@@ -269,13 +334,29 @@ hipError_t kf_ln_bwd(const void *dy, const void *x, const void *w,
     if (blocks > maxblk) blocks = maxblk;
     if (blocks < 1) blocks = 1;
     const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
+    static const int ilp_env = env_int("KF_LN_ILP", 2);
+    const bool ilp2 = ilp_env >= 2;
 #define CASE(KK)                                                            \
-    hipLaunchKernelGGL((ln_bwd_kernel<KK>), dim3((uint32_t)blocks),         \
-                       dim3(BLOCK), 2 * H * sizeof(float),                  \
-                       (hipStream_t)stream, (const unsigned short *)dy,     \
-                       (const unsigned short *)x, (const float *)w,         \
-                       (const float *)save_mean, (const float *)save_rstd,  \
-                       N, H, (unsigned short *)dx, (float *)wb_sums)
+    do {                                                                    \
+        if (ilp2)                                                           \
+            hipLaunchKernelGGL((ln_bwd_kernel<KK, 2>),                      \
+                               dim3((uint32_t)blocks), dim3(BLOCK),         \
+                               2 * H * sizeof(float), (hipStream_t)stream,  \
+                               (const unsigned short *)dy,                  \
+                               (const unsigned short *)x,                   \
+                               (const float *)w, (const float *)save_mean,  \
+                               (const float *)save_rstd, N, H,              \
+                               (unsigned short *)dx, (float *)wb_sums);     \
+        else                                                                \
+            hipLaunchKernelGGL((ln_bwd_kernel<KK, 1>),                      \
+                               dim3((uint32_t)blocks), dim3(BLOCK),         \
+                               2 * H * sizeof(float), (hipStream_t)stream,  \
+                               (const unsigned short *)dy,                  \
+                               (const unsigned short *)x,                   \
+                               (const float *)w, (const float *)save_mean,  \
+                               (const float *)save_rstd, N, H,              \
+                               (unsigned short *)dx, (float *)wb_sums);     \
+    } while (0)
     switch (K) {
     case 1: CASE(1); break;
     case 2: CASE(2); break;
