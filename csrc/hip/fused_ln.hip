@@ -334,7 +334,7 @@ hipError_t kf_ln_bwd(const void *dy, const void *x, const void *w,
     if (blocks > maxblk) blocks = maxblk;
     if (blocks < 1) blocks = 1;
     const int K = (H + WAVE * 8 - 1) / (WAVE * 8);
-    static const int ilp_env = env_int("KF_LN_ILP", 2);
+    static const int ilp_env = env_int("KF_LN_ILP", 1);  // 2 measured -5% (VGPR pressure)
     const bool ilp2 = ilp_env >= 2;
 #define CASE(KK)                                                            \
     do {                                                                    \
