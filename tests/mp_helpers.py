@@ -995,3 +995,37 @@ def monitored_all_reduce_body(rank, np):
     assert float(t2[0]) == 2.0 * np
     kf.finalize()
     return True
+
+
+def rccl_cpu_multihost_scopes_body(rank, np):
+    """Scope topology on a simulated 2-host cluster (loopback aliases):
+    LOCAL covers this host's ranks, CROSS the two local masters; the
+    capsule's host_rank/local_rank drive the (CPU-mode) controllers."""
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+    from kungfu_amd.ops import rccl
+
+    kf.init(with_torch=False)
+    rccl.init_cpu()
+    hosts = 2
+    my_host = rank % hosts
+    local_n = len([r for r in range(np) if r % hosts == my_host])
+    assert rccl.scope_size(rccl.LOCAL) == local_n
+    assert rccl.scope_size(rccl.CROSS) == hosts
+    is_master = rccl.scope_member(rccl.CROSS)
+    # local master = lowest global rank on this host = rank < hosts
+    assert is_master == (rank < hosts), (rank, is_master)
+    if is_master:
+        assert rccl.scope_rank(rccl.CROSS) == my_host
+    # LOCAL-scope ordering round executes on every host independently
+    from kungfu_amd import _rccl
+
+    got = []
+    rccl.scheduler_reset(["p", "q"], scope=rccl.LOCAL)
+    _rccl.start_task(rccl.LOCAL, "q", lambda: got.append("q"))
+    _rccl.start_task(rccl.LOCAL, "p", lambda: got.append("p"))
+    _rccl.drain(rccl.LOCAL)
+    assert got == ["p", "q"]
+    rccl.finalize()
+    kf.finalize()
+    return True

@@ -81,3 +81,17 @@ def _run_elastic(port_block):
 
 def test_rccl_elastic_reinit_cpu(port_block):
     _run_elastic(port_block)
+
+
+@retry_flaky
+def _run_multihost_scopes(port_block):
+    from tests.mp_helpers import (rccl_cpu_multihost_scopes_body,
+                                  spawn_multihost)
+
+    res = spawn_multihost(rccl_cpu_multihost_scopes_body, 4, port_block,
+                          hosts=2)
+    assert res == [True] * 4
+
+
+def test_rccl_multihost_scopes_cpu(port_block):
+    _run_multihost_scopes(port_block)
