@@ -15,8 +15,6 @@ hierarchical all-reduce without any torch.distributed sub-groups.
 On a GPU host the native layer is the DEFAULT GPU path; torch.distributed
 is only used when KUNGFU_GPU_BACKEND=torch is exported (fallback/debug).
 """
-import os
-
 import torch
 
 from kungfu_amd import _core
