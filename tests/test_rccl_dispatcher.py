@@ -95,3 +95,59 @@ def _run_multihost_scopes(port_block):
 
 def test_rccl_multihost_scopes_cpu(port_block):
     _run_multihost_scopes(port_block)
+
+
+def test_dispatcher_fuzz_single(port_block):
+    """Fuzz rounds in one process (cluster of 1): for random task-set
+    sizes and arrival permutations, release always follows the installed
+    order, and agree() adopts the recorded arrival order."""
+    import random
+    import subprocess
+    import sys
+
+    code = r"""
+import random
+import kungfu_amd as kf
+from kungfu_amd import _rccl
+from kungfu_amd.ops import rccl
+
+kf.init(with_torch=False)
+rccl.init_cpu()
+rng = random.Random(7)
+for trial in range(30):
+    n = rng.randrange(1, 9)
+    names = ["t%d" % i for i in range(n)]
+    rccl.scheduler_reset(names)
+    arrival = list(range(n))
+    rng.shuffle(arrival)
+    got = []
+    for slot in arrival:
+        _rccl.start_task(0, names[slot], (lambda s=slot: got.append(s)))
+    _rccl.drain(0)
+    assert got == list(range(n)), (trial, got)
+    assert rccl.last_arrival() == arrival
+    agreed = rccl.scheduler_agree()
+    assert agreed == arrival, (trial, agreed, arrival)
+    got2 = []
+    arrival2 = list(range(n))
+    rng.shuffle(arrival2)
+    for slot in arrival2:
+        _rccl.start_task(0, names[slot], (lambda s=slot: got2.append(s)))
+    _rccl.drain(0)
+    assert got2 == arrival, (trial, got2, arrival)
+rccl.finalize()
+kf.finalize()
+print("FUZZ OK")
+"""
+    import os
+
+    ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT
+    for k in ("KUNGFU_SELF_SPEC", "KUNGFU_INIT_PEERS", "RANK",
+              "WORLD_SIZE"):
+        env.pop(k, None)
+    out = subprocess.run([sys.executable, "-c", code], env=env, cwd=ROOT,
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0 and "FUZZ OK" in out.stdout, \
+        out.stdout + out.stderr
