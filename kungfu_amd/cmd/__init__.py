@@ -66,22 +66,28 @@ def run(argv=None):
     main(argv)
 
 
-def launch_multiprocess(fn, np, *args):
+def _mp_worker(fn, spec, peers, args):
+    # module-level so the spawn context can pickle it (a local closure
+    # cannot be); fn itself must be a module-level callable
+    os.environ["KUNGFU_SELF_SPEC"] = spec
+    os.environ["KUNGFU_INIT_PEERS"] = peers
+    fn(*args)
+
+
+def launch_multiprocess(fn, np, *args, port_base=34100):
     """Single-machine multiprocessing helper (reference
     launch_multiprocess): runs fn in np local worker processes via the
-    launcher env protocol using multiprocessing."""
+    launcher env protocol using multiprocessing (spawn)."""
     import multiprocessing as mp
 
     ctx = mp.get_context("spawn")
-    base = 34100
-    peers = ",".join("127.0.0.1:%d" % (base + i) for i in range(np))
-
-    def worker(i):
-        os.environ["KUNGFU_SELF_SPEC"] = "127.0.0.1:%d" % (base + i)
-        os.environ["KUNGFU_INIT_PEERS"] = peers
-        fn(*args)
-
-    procs = [ctx.Process(target=worker, args=(i,)) for i in range(np)]
+    peers = ",".join("127.0.0.1:%d" % (port_base + i) for i in range(np))
+    procs = [
+        ctx.Process(target=_mp_worker,
+                    args=(fn, "127.0.0.1:%d" % (port_base + i), peers,
+                          args))
+        for i in range(np)
+    ]
     for p in procs:
         p.start()
     for p in procs:

@@ -83,3 +83,38 @@ def test_rrun_dry_run():
     assert "10.0.0.2:30100" in lines[1]
     assert "KUNGFU_INIT_PEERS" in lines[0]
     assert "CUDA_VISIBLE_DEVICES=1" in lines[2]
+
+
+def test_launch_multiprocess_helper(port_block, tmp_path):
+    """kungfu_amd.cmd.launch_multiprocess (reference launch_multiprocess):
+    runs fn in np local processes over the env protocol. (A file-based
+    script: the spawn context must re-import __main__ to unpickle fn.)"""
+    import subprocess
+    import sys
+
+    code = """
+from kungfu_amd.cmd import launch_multiprocess
+
+def work():
+    import numpy as np
+    import kungfu_amd as kf
+    from kungfu_amd import _core
+    kf.init(with_torch=False)
+    a = np.ones(64, dtype=np.float32)
+    _core.all_reduce(a.ctypes.data, a.ctypes.data, a.size, 10, 0, "lm")
+    assert a[0] == 2.0, a[0]
+    kf.finalize()
+
+if __name__ == "__main__":
+    codes = launch_multiprocess(work, 2, port_base=%d)
+    assert codes == [0, 0], codes
+    print("LMP OK")
+""" % port_block
+    script = tmp_path / "lmp.py"
+    script.write_text(code)
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT
+    out = subprocess.run([sys.executable, str(script)], env=env, cwd=ROOT,
+                         capture_output=True, text=True, timeout=180)
+    assert out.returncode == 0 and "LMP OK" in out.stdout, \
+        out.stdout + out.stderr
