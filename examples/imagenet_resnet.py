@@ -82,7 +82,8 @@ def main():
                         step=trainer.step)
         monitor_epoch_end()
         print("epoch=%d step=%d size=%d loss=%.3f" %
-              (epoch, trainer.step, kf.size(), float(loss)), flush=True)
+              (epoch, trainer.step, kf.size(), float(loss.detach())),
+                  flush=True)
     monitor_train_end()
     print("DONE rank=%d steps=%d" % (kf.rank(), trainer.step), flush=True)
     kf.finalize()

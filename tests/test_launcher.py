@@ -118,3 +118,26 @@ if __name__ == "__main__":
                          capture_output=True, text=True, timeout=180)
     assert out.returncode == 0 and "LMP OK" in out.stdout, \
         out.stdout + out.stderr
+
+
+@retry_flaky
+def test_imagenet_resnet_example(port_block):
+    """The flagship example (elastic trainer + heartbeats + checkpoint
+    wiring + GNS probe) runs end-to-end on a tiny CPU config."""
+    import re
+    import subprocess
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    from mp_helpers import run_launcher_graceful
+
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "2",
+         "-port", str(port_block), "-port-range", str(port_block + 1),
+         sys.executable, "examples/imagenet_resnet.py", "--epochs", "1",
+         "--batch-size", "2", "--samples", "8", "--image-size", "64"],
+        ROOT, env, 280)
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
+    assert rc == 0, plain + err
+    assert plain.count("DONE rank=") == 2, plain
