@@ -75,6 +75,36 @@ def parse_args(argv=None):
     return args
 
 
+def parse_hostfile(text):
+    """MPI-style hostfile (reference plan/hostfile/hostfile.go): one
+    'ip [slots=N] [public_addr=X]' per line, #-comments; bare 'ip:slots'
+    lines are passed through."""
+    specs = []
+    for line in text.splitlines():
+        line = line.split("#", 1)[0].strip()
+        if not line:
+            continue
+        parts = line.split()
+        ip = parts[0]
+        if ":" in ip and len(parts) == 1:
+            specs.append(ip)  # already an ip:slots[:pub] spec
+            continue
+        slots, pub = 1, None
+        for kv in parts[1:]:
+            k, _, v = kv.partition("=")
+            if k == "slots":
+                slots = int(v)
+            elif k == "public_addr":
+                pub = v
+            else:
+                raise ValueError("bad hostfile entry: %r" % line)
+        spec = "%s:%d" % (ip, slots)
+        if pub:
+            spec += ":" + pub
+        specs.append(spec)
+    return ",".join(specs)
+
+
 def parse_duration(s):
     if s is None:
         return None
@@ -116,8 +146,7 @@ class Runner:
             args.self_ip = infer_self_ip(args.nic)
         if args.hostfile:
             with open(args.hostfile) as f:
-                hosts = ",".join(ln.strip() for ln in f
-                                 if ln.strip() and not ln.startswith("#"))
+                hosts = parse_hostfile(f.read())
         elif args.hosts:
             hosts = args.hosts
         else:

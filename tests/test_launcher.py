@@ -141,3 +141,22 @@ def test_imagenet_resnet_example(port_block):
     plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
     assert rc == 0, plain + err
     assert plain.count("DONE rank=") == 2, plain
+
+
+def test_hostfile_mpi_format(tmp_path):
+    """-hostfile accepts the MPI-style format the reference parses
+    (plan/hostfile/hostfile.go): 'ip slots=N [public_addr=X]' plus
+    comments; bare ip:slots lines pass through."""
+    from kungfu_amd.launcher.run import parse_hostfile
+
+    text = ("# cluster\n"
+            "10.0.0.1 slots=4\n"
+            "10.0.0.2 slots=2 public_addr=1.2.3.4  # edge\n"
+            "10.0.0.3\n"
+            "10.0.0.4:8\n")
+    assert parse_hostfile(text) == \
+        "10.0.0.1:4,10.0.0.2:2:1.2.3.4,10.0.0.3:1,10.0.0.4:8"
+    from kungfu_amd import _core
+
+    peers = _core.gen_peer_list(parse_hostfile(text), 7, 30000)
+    assert len(peers.split(",")) == 7
