@@ -212,3 +212,38 @@ def test_checkpoint_roundtrip(tmp_path):
         o.step()
     for p1, p2 in zip(model.parameters(), model2.parameters()):
         assert torch.equal(p1, p2)
+
+
+def test_configserver_http_methods(port_block):
+    """GET/PUT/POST/DELETE of the elastic config server (reference
+    elastic/configserver: version counter, clear semantics)."""
+    import json
+    import urllib.request
+
+    import threading
+
+    from kungfu_amd.launcher.configserver import make_server
+
+    port = port_block + 70
+    srv = make_server(port, host="127.0.0.1")
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        url = "http://127.0.0.1:%d/" % port
+        body = json.dumps({"workers": ["127.0.0.1:1"]}).encode()
+        for method in ("PUT", "POST"):
+            req = urllib.request.Request(url, data=body, method=method)
+            assert urllib.request.urlopen(req, timeout=5).status == 200
+        got = json.loads(urllib.request.urlopen(url, timeout=5).read())
+        assert got.get("workers") == ["127.0.0.1:1"]
+        req = urllib.request.Request(url, method="DELETE")
+        assert urllib.request.urlopen(req, timeout=5).status == 200
+        import urllib.error
+
+        try:
+            resp = urllib.request.urlopen(url, timeout=5)
+            cleared = resp.status != 200 or not resp.read()
+        except urllib.error.HTTPError:
+            cleared = True
+        assert cleared
+    finally:
+        srv.shutdown()
