@@ -225,7 +225,7 @@ def test_configserver_http_methods(port_block):
     from kungfu_amd.launcher.configserver import make_server
 
     port = port_block + 70
-    srv = make_server(port, host="127.0.0.1")
+    srv, _state = make_server(port, host="127.0.0.1")
     threading.Thread(target=srv.serve_forever, daemon=True).start()
     try:
         url = "http://127.0.0.1:%d/" % port
