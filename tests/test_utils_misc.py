@@ -220,13 +220,12 @@ def test_configserver_http_methods(port_block):
     import json
     import urllib.request
 
-    import threading
-
     from kungfu_amd.launcher.configserver import make_server
 
     port = port_block + 70
-    srv, _state = make_server(port, host="127.0.0.1")
-    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    # make_server returns (srv, thread) with the thread NOT yet started
+    srv, thread = make_server(port, host="127.0.0.1")
+    thread.start()
     try:
         url = "http://127.0.0.1:%d/" % port
         body = json.dumps({"workers": ["127.0.0.1:1"]}).encode()
