@@ -285,7 +285,9 @@ def metrics_body(rank, np):
         "http://127.0.0.1:%d/metrics" % port, timeout=5).read().decode()
     kf.barrier()
     kf.finalize()
-    return "kungfu_egress_bytes_total" in body
+    # both directions must be metered (round-2: ingress added)
+    return ("kungfu_egress_bytes_total" in body and
+            "kungfu_ingress_bytes_total" in body)
 
 
 def sampler_body(rank, np):
