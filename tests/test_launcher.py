@@ -160,3 +160,27 @@ def test_hostfile_mpi_format(tmp_path):
 
     peers = _core.gen_peer_list(parse_hostfile(text), 7, 30000)
     assert len(peers.split(",")) == 7
+
+
+@retry_flaky
+def test_allreduce_microbenchmark_cpu(port_block):
+    """`python -m kungfu_amd.benchmarks --method CPU` under kungfu-run
+    (reference `python -m kungfu.tensorflow.v1.benchmarks`): emits the
+    algbw report line."""
+    import re
+    import sys
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    from mp_helpers import run_launcher_graceful
+
+    rc, out, err = run_launcher_graceful(
+        [sys.executable, "-m", "kungfu_amd.run", "-np", "2",
+         "-port", str(port_block), "-port-range", str(port_block + 1),
+         sys.executable, "-m", "kungfu_amd.benchmarks", "--method", "CPU",
+         "--model", "slp-mnist", "--steps", "5", "--warmup", "2"],
+        ROOT, env, 200)
+    plain = re.sub(r"\x1b\[[0-9;]*m", "", out)
+    assert rc == 0, plain + err
+    m = re.search(r"method=CPU .*np=2 .*algbw=([0-9.]+) GB/s", plain)
+    assert m and float(m.group(1)) > 0, plain
